@@ -1,0 +1,798 @@
+"""Connection slot management: SocketMgrFSM, ClaimHandle, ConnectionSlotFSM.
+
+This is the guts of connection management (reference
+lib/connection-fsm.js).  Three cooperating Moore machines:
+
+- **SocketMgrFSM** — owns the actual connection object produced by the
+  user-supplied ``constructor(backend)``; handles connect timeouts,
+  error/close de-duplication, exponential backoff with jitter, and
+  "monitor mode" (infinite retries at max backoff, watching a dead
+  backend; lib/connection-fsm.js:68-420).
+  States: init -> connecting -> connected -> {error, closed} -> backoff
+  -> {connecting, failed}.
+
+- **ClaimHandle** — the object handed to pool users; encapsulates the
+  claim double-handshake (try -> claim -> accept/reject), the claim
+  timeout, cancellation, and the event-listener leak detector
+  (lib/connection-fsm.js:422-808).
+  States: waiting -> claiming -> claimed -> {released, closed}, plus
+  cancelled and failed.
+
+- **ConnectionSlotFSM** — one per pool/set slot; drives the SocketMgr,
+  decides when to retry or give up, and reports only the transitions the
+  Pool/Set cares about (lib/connection-fsm.js:810-1242).
+  States: init -> connecting -> {failed, retrying, idle} ;
+  idle <-> busy ; killing/stopping/stopped.
+
+The user Connection interface is the reference's contract
+(docs/api.adoc:580-645): an EventEmitter emitting ``connect``, ``close``,
+``error`` (and optionally ``timeout``/``connectError``/``connectTimeout``)
+with a ``destroy()`` method and optional ``set_unwanted()``/``ref()``/
+``unref()``.
+"""
+
+from __future__ import annotations
+
+import math
+from typing import Any, Callable, Dict, List, Optional
+
+from . import errors as mod_errors
+from . import utils as mod_utils
+from .events import EventEmitter
+from .fsm import FSM, FSMError, StateScope
+from .logutil import CueballLogger
+
+__all__ = ["SocketMgrFSM", "ClaimHandle", "ConnectionSlotFSM"]
+
+
+def _call_optional(obj: Any, *names: str) -> None:
+    for name in names:
+        fn = getattr(obj, name, None)
+        if callable(fn):
+            fn()
+            return
+
+
+class SocketMgrFSM(FSM):
+    """Socket lifecycle + retry/backoff engine (1:1 with a slot FSM)."""
+
+    def __init__(self, options: Dict[str, Any]) -> None:
+        recovery = options["recovery"]
+        connect_recov = recovery["default"]
+        initial_recov = recovery["default"]
+        if recovery.get("connect") is not None:
+            initial_recov = recovery["connect"]
+            connect_recov = recovery["connect"]
+        if recovery.get("initial") is not None:
+            initial_recov = recovery["initial"]
+        mod_utils.assert_recovery(connect_recov, "recovery.connect")
+        mod_utils.assert_recovery(initial_recov, "recovery.initial")
+
+        self.sm_initial_recov = initial_recov
+        self.sm_connect_recov = connect_recov
+        self.sm_pool = options["pool"]
+        self.sm_backend = options["backend"]
+        self.sm_constructor = options["constructor"]
+        self.sm_slot = options["slot"]
+        log: CueballLogger = options["log"]
+        self.sm_log = log.child(
+            component="SocketMgrFSM",
+            backend=self.sm_backend.get("key"),
+            address=self.sm_backend.get("address"),
+            port=self.sm_backend.get("port"),
+        )
+        self.sm_last_error: Optional[BaseException] = None
+        self.sm_socket: Optional[EventEmitter] = None
+        self.sm_monitor: Optional[bool] = None
+        # backoff bookkeeping, filled by reset_backoff()
+        self.sm_retries = 0.0
+        self.sm_retries_left = 0.0
+        self.sm_min_delay = 0.0
+        self.sm_delay = 0.0
+        self.sm_max_delay = math.inf
+        self.sm_timeout = 0.0
+        self.sm_max_timeout = math.inf
+        self.sm_delay_spread = 0.2
+
+        super().__init__("init", loop=options.get("loop"))
+        self.set_monitor(bool(options["monitor"]))
+
+    # -- monitor mode / backoff ---------------------------------------
+    def set_monitor(self, value: bool) -> None:
+        """In monitor mode: infinite retries, delay/timeout pinned at max
+        (lib/connection-fsm.js:175-208)."""
+        if not (self.is_in_state("init") or self.is_in_state("connected")):
+            raise FSMError("set_monitor only valid in init/connected")
+        if value == self.sm_monitor:
+            return
+        self.sm_monitor = value
+        self.reset_backoff()
+
+    def reset_backoff(self) -> None:
+        recov = self.sm_initial_recov
+        self.sm_retries = recov["retries"]
+        self.sm_retries_left = recov["retries"]
+        self.sm_min_delay = recov["delay"]
+        self.sm_delay = recov["delay"]
+        self.sm_max_delay = recov.get("maxDelay") or math.inf
+        self.sm_timeout = recov["timeout"]
+        self.sm_max_timeout = recov.get("maxTimeout") or math.inf
+        self.sm_delay_spread = recov.get("delaySpread") or 0.2
+
+        if self.sm_monitor is True:
+            mult = 1 << int(self.sm_retries)
+            self.sm_delay = self.sm_max_delay
+            if not math.isfinite(self.sm_delay):
+                self.sm_delay = recov["delay"] * mult
+            self.sm_timeout = self.sm_max_timeout
+            if not math.isfinite(self.sm_timeout):
+                self.sm_timeout = recov["timeout"] * mult
+            # keep retrying a failed backend forever
+            self.sm_retries = math.inf
+            self.sm_retries_left = math.inf
+
+    def set_unwanted(self) -> None:
+        """Forward to the connection if it supports it (advisory only;
+        lib/connection-fsm.js:210-221)."""
+        if self.sm_socket is not None:
+            _call_optional(self.sm_socket, "set_unwanted", "setUnwanted")
+
+    # -- signal functions ---------------------------------------------
+    def connect(self) -> None:
+        if not (self.is_in_state("init") or self.is_in_state("closed")):
+            raise FSMError('SocketMgrFSM.connect only in "init"/"closed" '
+                           '(is in "%s")' % self.get_state())
+        self.emit("connectAsserted")
+
+    def retry(self) -> None:
+        if not (self.is_in_state("closed") or self.is_in_state("error")):
+            raise FSMError('SocketMgrFSM.retry only in "closed"/"error" '
+                           '(is in "%s")' % self.get_state())
+        self.emit("retryAsserted")
+
+    def close(self) -> None:
+        if not (self.is_in_state("connected") or self.is_in_state("backoff")):
+            raise FSMError('SocketMgrFSM.close only in "connected"/"backoff" '
+                           '(is in "%s")' % self.get_state())
+        self.emit("closeAsserted")
+
+    def get_last_error(self) -> Optional[BaseException]:
+        return self.sm_last_error
+
+    def get_socket(self) -> Any:
+        if not self.is_in_state("connected"):
+            raise FSMError("get_socket only in connected state")
+        return self.sm_socket
+
+    # -- states --------------------------------------------------------
+    def state_init(self, S: StateScope) -> None:
+        S.valid_transitions(["connecting"])
+        S.on(self, "connectAsserted", lambda: S.goto_state("connecting"))
+
+    def state_connecting(self, S: StateScope) -> None:
+        S.valid_transitions(["connected", "error"])
+
+        def on_timeout() -> None:
+            self.sm_last_error = mod_errors.ConnectionTimeoutError(
+                self.sm_backend)
+            S.goto_state("error")
+            self.sm_pool._incr_counter("timeout-during-connect")
+
+        S.timeout(self.sm_timeout, on_timeout)
+
+        self.sm_log.trace("calling constructor to open new connection")
+        sock = self.sm_constructor(self.sm_backend)
+        if sock is None:
+            raise TypeError("connection constructor returned None")
+        self.sm_socket = sock
+        sock.sm_fsm = self
+
+        S.on(sock, "connect", lambda: S.goto_state("connected"))
+
+        def on_error(err: BaseException) -> None:
+            self.sm_last_error = mod_errors.ConnectionError_(
+                self.sm_backend, "error", "connect", err)
+            S.goto_state("error")
+            self.sm_log.trace("emitted error while connecting")
+            self.sm_pool._incr_counter("error-during-connect")
+
+        on_error._cueball_internal = True  # type: ignore[attr-defined]
+        S.on(sock, "error", on_error)
+
+        def on_connect_error(err: BaseException) -> None:
+            self.sm_last_error = mod_errors.ConnectionError_(
+                self.sm_backend, "connectError", "connect", err)
+            S.goto_state("error")
+            self.sm_log.trace("emitted connectError while connecting")
+            self.sm_pool._incr_counter("error-during-connect")
+
+        S.on(sock, "connectError", on_connect_error)
+
+        def on_close() -> None:
+            self.sm_last_error = mod_errors.ConnectionClosedError(
+                self.sm_backend)
+            S.goto_state("error")
+            self.sm_log.trace("closed while connecting")
+            self.sm_pool._incr_counter("close-during-connect")
+
+        S.on(sock, "close", on_close)
+
+        def on_sock_timeout() -> None:
+            self.sm_last_error = mod_errors.ConnectionTimeoutError(
+                self.sm_backend)
+            S.goto_state("error")
+            self.sm_log.trace("timed out while connecting")
+            self.sm_pool._incr_counter("timeout-during-connect")
+
+        S.on(sock, "timeout", on_sock_timeout)
+        S.on(sock, "connectTimeout", on_sock_timeout)
+
+    def state_connected(self, S: StateScope) -> None:
+        S.valid_transitions(["error", "closed"])
+        sock = self.sm_socket
+        self.sm_log.trace("connected")
+        self.reset_backoff()
+
+        def on_error(err: BaseException) -> None:
+            self.sm_last_error = mod_errors.ConnectionError_(
+                self.sm_backend, "error", "operation", err)
+            S.goto_state("error")
+            self.sm_pool._incr_counter("error-while-connected")
+            self.sm_log.trace("emitted error while connected")
+
+        on_error._cueball_internal = True  # type: ignore[attr-defined]
+        S.on(sock, "error", on_error)
+        S.on(sock, "close", lambda: S.goto_state("closed"))
+        S.on(self, "closeAsserted", lambda: S.goto_state("closed"))
+
+    def state_error(self, S: StateScope) -> None:
+        S.valid_transitions(["backoff"])
+        if self.sm_socket is not None:
+            self.sm_socket.destroy()
+        self.sm_socket = None
+        S.on(self, "retryAsserted", lambda: S.goto_state("backoff"))
+
+    def state_backoff(self, S: StateScope) -> None:
+        S.valid_transitions(["failed", "connecting", "closed"])
+
+        # NB: "retries" means "attempts" in the cueball API, so compare
+        # to 1 rather than 0 (lib/connection-fsm.js:364-372).
+        if self.sm_retries_left != math.inf and self.sm_retries_left <= 1:
+            S.goto_state("failed")
+            return
+
+        delay = mod_utils.gen_delay(self.sm_delay, self.sm_delay_spread)
+
+        if self.sm_retries != math.inf:
+            self.sm_retries_left -= 1
+            self.sm_delay *= 2
+            self.sm_timeout *= 2
+            if self.sm_timeout > self.sm_max_timeout:
+                self.sm_timeout = self.sm_max_timeout
+            if self.sm_delay > self.sm_max_delay:
+                self.sm_delay = self.sm_max_delay
+
+        S.timeout(delay, lambda: S.goto_state("connecting"))
+        S.on(self, "closeAsserted", lambda: S.goto_state("closed"))
+
+    def state_closed(self, S: StateScope) -> None:
+        S.valid_transitions(["backoff", "connecting"])
+        if self.sm_socket is not None:
+            self.sm_socket.destroy()
+        self.sm_socket = None
+        self.sm_log.trace("connection closed")
+        S.on(self, "retryAsserted", lambda: S.goto_state("backoff"))
+        S.on(self, "connectAsserted", lambda: S.goto_state("connecting"))
+
+    def state_failed(self, S: StateScope) -> None:
+        S.valid_transitions([])
+        self.sm_log.warn("failed to connect to backend, retries exhausted")
+        self.sm_pool._incr_counter("retries-exhausted")
+
+
+class ClaimHandle(FSM):
+    """Handle given to pool users; drives the claim double-handshake.
+
+    Reference: CueBallClaimHandle, lib/connection-fsm.js:422-784.
+    """
+
+    # events a connection claimer may leak handlers for
+    _LEAK_EVENTS = ("close", "error", "readable", "data")
+
+    def __init__(self, options: Dict[str, Any]) -> None:
+        self.ch_claim_timeout = options["claimTimeout"]
+        self.ch_pool = options["pool"]
+        throw_error = options.get("throwError")
+        self.ch_throw_error = True if throw_error is None else bool(throw_error)
+        self.ch_claim_stack: List[str] = list(options["claimStack"])
+        self.ch_callback: Callable = options["callback"]
+        log: CueballLogger = options["log"]
+        self.ch_log = log.child(component="ClaimHandle")
+
+        self.ch_slot: Optional["ConnectionSlotFSM"] = None
+        self.ch_release_stack: Optional[List[str]] = None
+        self.ch_connection: Any = None
+        self.ch_pre_listeners: Dict[str, int] = {}
+        self.ch_cancelled = False
+        self.ch_last_error: Optional[BaseException] = None
+        self.ch_do_release_leak_check = True
+        self.ch_pinger = False
+
+        super().__init__("waiting", loop=options.get("loop"))
+        self.ch_started = self._loop.time() * 1000.0
+
+    # -- misuse traps (lib/connection-fsm.js:529-557) -------------------
+    @property
+    def writable(self) -> Any:
+        raise mod_errors.ClaimHandleMisusedError()
+
+    @property
+    def readable(self) -> Any:
+        raise mod_errors.ClaimHandleMisusedError()
+
+    def on(self, event: str, listener: Callable) -> Callable:
+        if event in ("readable", "close"):
+            raise mod_errors.ClaimHandleMisusedError()
+        return super().on(event, listener)
+
+    def once(self, event: str, listener: Callable) -> Callable:
+        if event in ("readable", "close"):
+            raise mod_errors.ClaimHandleMisusedError()
+        return super().once(event, listener)
+
+    def disable_release_leak_check(self) -> None:
+        self.ch_do_release_leak_check = False
+
+    # -- signal functions ----------------------------------------------
+    def try_(self, slot: "ConnectionSlotFSM") -> None:
+        """The pool offers `slot` to this handle (must be idle)."""
+        if not self.is_in_state("waiting"):
+            raise FSMError('ClaimHandle.try_ only in "waiting" (is in "%s")'
+                           % self.get_state())
+        if not slot.is_in_state("idle"):
+            raise FSMError('ClaimHandle.try_ needs an idle slot (is in "%s")'
+                           % slot.get_state())
+        self.ch_slot = slot
+        self.emit("tryAsserted")
+
+    def accept(self, connection: Any) -> None:
+        if not self.is_in_state("claiming"):
+            raise FSMError("accept only in claiming")
+        self.ch_connection = connection
+        self.emit("accepted")
+
+    def reject(self) -> None:
+        if not self.is_in_state("claiming"):
+            raise FSMError("reject only in claiming")
+        self.emit("rejected")
+
+    def cancel(self) -> None:
+        if self.is_in_state("claimed"):
+            self.release()
+        else:
+            self.ch_cancelled = True
+            self.emit("cancelled")
+
+    def timeout(self) -> None:
+        if not self.is_in_state("waiting"):
+            raise FSMError("timeout only in waiting")
+        self.emit("timeout")
+
+    def fail(self, err: BaseException) -> None:
+        self.emit("error", err)
+
+    def _relinquish(self, event: str) -> None:
+        if not self.is_in_state("claimed"):
+            if self.is_in_state("released") or self.is_in_state("closed"):
+                stack = self.ch_release_stack or ["?", "?", "?"]
+                by = stack[2] if len(stack) > 2 else stack[-1]
+                raise mod_errors.CueballError(
+                    "Connection not claimed by this handle, released by %s"
+                    % by)
+            raise mod_errors.CueballError(
+                'ClaimHandle.release() called while in state "%s"'
+                % self.get_state())
+        self.ch_release_stack = mod_utils.maybe_capture_stack_trace()
+        self.emit(event)
+
+    def release(self) -> None:
+        self._relinquish("releaseAsserted")
+
+    def close(self) -> None:
+        self._relinquish("closeAsserted")
+
+    # -- states ----------------------------------------------------------
+    def state_waiting(self, S: StateScope) -> None:
+        S.valid_transitions(["claiming", "cancelled", "failed"])
+        self.ch_slot = None
+
+        S.on(self, "tryAsserted", lambda: S.goto_state("claiming"))
+
+        def on_timeout() -> None:
+            self.ch_last_error = mod_errors.ClaimTimeoutError(self.ch_pool)
+            self.ch_pool._incr_counter("claim-timeout")
+            S.goto_state("failed")
+
+        if math.isfinite(self.ch_claim_timeout):
+            S.timeout(self.ch_claim_timeout, on_timeout)
+        S.on(self, "timeout", on_timeout)
+
+        def on_error(err: BaseException) -> None:
+            self.ch_last_error = err
+            S.goto_state("failed")
+
+        S.on(self, "error", on_error)
+        S.on(self, "cancelled", lambda: S.goto_state("cancelled"))
+
+    def state_claiming(self, S: StateScope) -> None:
+        S.valid_transitions(["claimed", "waiting", "cancelled"])
+
+        S.on(self, "accepted", lambda: S.goto_state("claimed"))
+
+        def on_rejected() -> None:
+            if self.ch_cancelled:
+                S.goto_state("cancelled")
+            else:
+                S.goto_state("waiting")
+
+        S.on(self, "rejected", on_rejected)
+        self.ch_slot.claim(self)
+
+    def state_claimed(self, S: StateScope) -> None:
+        S.valid_transitions(["released", "closed"])
+
+        S.on(self, "releaseAsserted", lambda: S.goto_state("released"))
+        S.on(self, "closeAsserted", lambda: S.goto_state("closed"))
+
+        if self.ch_cancelled:
+            S.goto_state("released")
+            return
+
+        conn = self.ch_connection
+        self.ch_pre_listeners = {}
+        for evt in self._LEAK_EVENTS:
+            self.ch_pre_listeners[evt] = count_listeners(conn, evt)
+
+        def on_conn_error(err: BaseException) -> None:
+            count = count_listeners(conn, "error")
+            if count == 0 and self.ch_throw_error:
+                # End-user never registered an 'error' listener: surface
+                # the failure loudly (lib/connection-fsm.js:697-706).
+                raise err
+            self.ch_log.warn(
+                "connection emitted error while claimed "
+                '(for claim callback "%s")',
+                getattr(self.ch_callback, "__name__", "?"))
+            self.ch_pool._incr_counter("error-while-claimed")
+
+        on_conn_error._cueball_internal = True  # type: ignore[attr-defined]
+        S.on(conn, "error", on_conn_error)
+
+        self.ch_callback(None, self, conn)
+
+    def state_released(self, S: StateScope) -> None:
+        S.valid_transitions([])
+        if not self.ch_do_release_leak_check:
+            return
+        conn = self.ch_connection
+        for evt in self._LEAK_EVENTS:
+            new_count = count_listeners(conn, evt)
+            old_count = self.ch_pre_listeners.get(evt)
+            if old_count is not None and new_count > old_count:
+                self.ch_log.warn(
+                    "connection claimer looks like it leaked event "
+                    "handlers", event=evt, count_before_claim=old_count,
+                    count_after_release=new_count)
+
+    def state_closed(self, S: StateScope) -> None:
+        S.valid_transitions([])
+        # no leak check: the connection is being torn down anyway
+
+    def state_cancelled(self, S: StateScope) -> None:
+        S.valid_transitions([])
+        # Public API: the callback must NOT be called after .cancel().
+
+    def state_failed(self, S: StateScope) -> None:
+        S.valid_transitions([])
+        S.immediate(lambda: self.ch_callback(self.ch_last_error))
+
+
+def count_listeners(emitter: Any, event: str) -> int:
+    """Count user-registered listeners, ignoring cueball's own internal
+    handlers (lib/connection-fsm.js:786-808)."""
+    if not hasattr(emitter, "listeners"):
+        return 0
+    n = 0
+    for h in emitter.listeners(event):
+        if not callable(h):
+            continue
+        target = getattr(h, "listener", h)
+        if getattr(h, "_cueball_internal", False) or \
+                getattr(target, "_cueball_internal", False):
+            continue
+        n += 1
+    return n
+
+
+class ConnectionSlotFSM(FSM):
+    """One pool/set slot: drives a SocketMgrFSM and reports the
+    transitions its owner cares about (lib/connection-fsm.js:810-1242)."""
+
+    def __init__(self, options: Dict[str, Any]) -> None:
+        self.csf_pool = options["pool"]
+        self.csf_backend = options["backend"]
+        self.csf_wanted = True
+        self.csf_handle: Optional[ClaimHandle] = None
+        self.csf_prev_handle: Optional[ClaimHandle] = None
+        self.csf_monitor = bool(options["monitor"])
+        self.csf_checker = options.get("checker")
+        self.csf_check_timeout = options.get("checkTimeout")
+        log: CueballLogger = options["log"]
+        self.csf_log = log.child(
+            component="ConnectionSlotFSM",
+            backend=self.csf_backend.get("key"),
+            address=self.csf_backend.get("address"),
+            port=self.csf_backend.get("port"),
+        )
+        self.csf_smgr = SocketMgrFSM({
+            "pool": options["pool"],
+            "constructor": options["constructor"],
+            "backend": options["backend"],
+            "log": options["log"],
+            "recovery": options["recovery"],
+            "monitor": bool(options["monitor"]),
+            "slot": self,
+            "loop": options.get("loop"),
+        })
+        super().__init__("init", loop=options.get("loop"))
+
+    # -- signal functions ----------------------------------------------
+    def set_unwanted(self) -> None:
+        if self.csf_wanted is False:
+            return
+        self.csf_wanted = False
+        self.csf_smgr.set_unwanted()
+        self.emit("unwanted")
+
+    def start(self) -> None:
+        if not self.is_in_state("init"):
+            raise FSMError("start only in init")
+        self.emit("startAsserted")
+
+    def claim(self, handle: ClaimHandle) -> None:
+        if not self.is_in_state("idle"):
+            raise FSMError("claim only in idle")
+        if self.csf_handle is not None:
+            raise FSMError("slot already has a handle")
+        self.csf_handle = handle
+        self.emit("claimAsserted")
+
+    def make_child_logger(self, **fields: Any) -> CueballLogger:
+        return self.csf_log.child(**fields)
+
+    def get_socket_mgr(self) -> SocketMgrFSM:
+        return self.csf_smgr
+
+    def get_backend(self) -> Dict[str, Any]:
+        return self.csf_backend
+
+    def is_running_ping(self) -> bool:
+        return bool(self.is_in_state("busy") and self.csf_handle is not None
+                    and self.csf_handle.ch_pinger)
+
+    # -- states ----------------------------------------------------------
+    def state_init(self, S: StateScope) -> None:
+        S.on(self, "startAsserted", lambda: S.goto_state("connecting"))
+
+    def state_connecting(self, S: StateScope) -> None:
+        S.valid_transitions(["failed", "retrying", "idle"])
+        smgr = self.csf_smgr
+
+        def on_smgr_state(st: str) -> None:
+            if st in ("init", "connecting"):
+                return
+            if st == "failed":
+                S.goto_state("failed")
+            elif st == "error":
+                S.goto_state("retrying")
+            elif st == "connected":
+                S.goto_state("idle")
+            else:
+                raise FSMError(
+                    'Unhandled smgr state transition: .connect() => "%s"'
+                    % st)
+
+        S.on(smgr, "stateChanged", on_smgr_state)
+        smgr.connect()
+
+    def state_failed(self, S: StateScope) -> None:
+        S.valid_transitions([])
+        if not self.csf_smgr.is_in_state("failed"):
+            raise FSMError("smgr must be failed")
+
+    def state_retrying(self, S: StateScope) -> None:
+        S.valid_transitions(["idle", "failed", "retrying", "stopped",
+                             "stopping"])
+        smgr = self.csf_smgr
+
+        def on_smgr_state(st: str) -> None:
+            if st in ("backoff", "connecting"):
+                return
+            if st == "failed":
+                S.goto_state("failed")
+            elif st == "error":
+                if self.csf_monitor and not self.csf_wanted:
+                    S.goto_state("stopped")
+                else:
+                    S.goto_state("retrying")
+            elif st == "connected":
+                S.goto_state("idle")
+            else:
+                raise FSMError(
+                    'Unhandled smgr state transition: .retry() => "%s"' % st)
+
+        S.on(smgr, "stateChanged", on_smgr_state)
+
+        def on_unwanted() -> None:
+            if self.csf_monitor and smgr.is_in_state("backoff"):
+                S.goto_state("stopping")
+
+        S.on(self, "unwanted", on_unwanted)
+        smgr.retry()
+
+    def state_idle(self, S: StateScope) -> None:
+        S.valid_transitions(["retrying", "connecting", "stopping", "stopped",
+                             "busy"])
+        smgr = self.csf_smgr
+
+        if self.csf_handle is not None:
+            self.csf_prev_handle = self.csf_handle
+        self.csf_handle = None
+
+        # A monitor that connected becomes a normal slot
+        # (lib/connection-fsm.js:1053-1057).
+        if self.csf_monitor is True:
+            self.csf_monitor = False
+            smgr.set_monitor(False)
+
+        def on_unwanted() -> None:
+            if smgr.is_in_state("connected"):
+                S.goto_state("stopping")
+
+        if not self.csf_wanted:
+            on_unwanted()
+            return
+        S.on(self, "unwanted", on_unwanted)
+
+        def on_smgr_state(st: str) -> None:
+            if st == "error":
+                S.goto_state("retrying")
+            elif st == "closed":
+                if not self.csf_wanted:
+                    S.goto_state("stopped")
+                else:
+                    S.goto_state("connecting")
+            else:
+                raise FSMError(
+                    'Unhandled smgr state transition: connected => "%s"' % st)
+
+        S.on(smgr, "stateChanged", on_smgr_state)
+        S.on(self, "claimAsserted", lambda: S.goto_state("busy"))
+
+        if self.csf_check_timeout is not None and \
+                self.csf_checker is not None:
+            S.timeout(self.csf_check_timeout,
+                      lambda: _do_ping_check(self, self.csf_checker))
+
+    def state_busy(self, S: StateScope) -> None:
+        S.valid_transitions(["idle", "stopping", "stopped", "retrying",
+                             "killing", "connecting"])
+        smgr = self.csf_smgr
+        hdl = self.csf_handle
+        # Track the smgr state as *observed through events*: a transition
+        # that happened this same loop spin is still pending delivery, and
+        # the exit decision must match what we have actually seen
+        # (lib/connection-fsm.js:885-890, :1129-1196).
+        observed = {"st": "connected"}
+        S.on(smgr, "stateChanged", lambda st: observed.update(st=st))
+
+        def on_release() -> None:
+            st = observed["st"]
+            if st == "connected":
+                if self.csf_wanted:
+                    S.goto_state("idle")
+                else:
+                    S.goto_state("stopping")
+            elif st == "closed":
+                if self.csf_wanted:
+                    S.goto_state("connecting")
+                else:
+                    S.goto_state("stopped")
+            elif st == "error":
+                S.goto_state("retrying")
+            else:
+                raise FSMError(
+                    'Handle released while smgr was in unhandled state "%s"'
+                    % smgr.get_state())
+
+        def on_close() -> None:
+            if observed["st"] == "connected":
+                S.goto_state("killing")
+            else:
+                S.goto_state("retrying")
+
+        def on_hdl_state(st: str) -> None:
+            if st == "released":
+                on_release()
+            elif st == "closed":
+                on_close()
+
+        S.on(hdl, "stateChanged", on_hdl_state)
+
+        # The smgr may have left 'connected' before we got here; if we
+        # lost the race, treat it like our handle was released.
+        if smgr.is_in_state("connected"):
+            hdl.accept(smgr.get_socket())
+        else:
+            hdl.reject()
+            self.csf_handle = None
+            on_release()
+
+    def state_killing(self, S: StateScope) -> None:
+        S.valid_transitions(["retrying"])
+        smgr = self.csf_smgr
+
+        def on_smgr_state(st: str) -> None:
+            if st in ("closed", "error"):
+                S.goto_state("retrying")
+
+        S.on(smgr, "stateChanged", on_smgr_state)
+        # The socket may already be closed with the event still pending;
+        # if so just wait for it (lib/connection-fsm.js:1209-1216).
+        if not smgr.is_in_state("closed") and not smgr.is_in_state("error"):
+            smgr.close()
+
+    def state_stopping(self, S: StateScope) -> None:
+        S.valid_transitions(["stopped"])
+        smgr = self.csf_smgr
+
+        def on_smgr_state(st: str) -> None:
+            if st in ("closed", "error"):
+                S.goto_state("stopped")
+
+        S.on(smgr, "stateChanged", on_smgr_state)
+        if not smgr.is_in_state("closed") and not smgr.is_in_state("error"):
+            smgr.close()
+
+    def state_stopped(self, S: StateScope) -> None:
+        S.valid_transitions([])
+        smgr = self.csf_smgr
+        if not (smgr.is_in_state("closed") or smgr.is_in_state("error")
+                or smgr.is_in_state("failed")):
+            raise FSMError("smgr must be stopped")
+
+
+def _do_ping_check(fsm: ConnectionSlotFSM, checker: Callable) -> None:
+    """Health-check an idle slot: claim it internally and run the checker
+    (lib/connection-fsm.js:1101-1127)."""
+
+    def ping_check_adapter(err: Optional[BaseException], hdl: ClaimHandle,
+                           conn: Any = None) -> None:
+        # infinite timeout + no .fail() => err is always None here
+        if err is not None:
+            raise FSMError("ping check handle failed unexpectedly")
+        checker(hdl, conn)
+
+    handle = ClaimHandle({
+        "pool": fsm.csf_pool,
+        "claimStack": ["claim", "cueball._do_ping_check",
+                       "cueball._do_ping_check"],
+        "callback": ping_check_adapter,
+        "log": fsm.csf_log,
+        "claimTimeout": math.inf,
+        "loop": fsm._loop,
+    })
+    handle.ch_pinger = True
+    # If the try fails and the handle returns to "waiting", just let go
+    # of it entirely.
+    handle.try_(fsm)

@@ -1,0 +1,277 @@
+"""Moore finite-state-machine runtime.
+
+This is the foundation the entire framework is written against — the
+equivalent of the ``mooremachine`` npm module that every class in the
+reference extends (reference package.json:13; every lib/*.js file calls
+``FSM.call(this, <state>)``).  It is a fresh design for asyncio, but it
+reproduces the precise observable semantics the reference's correctness
+depends on (survey §7 "hard parts" #1):
+
+1. **State-entry functions.**  A subclass defines ``state_<name>(self, S)``
+   methods.  Dotted sub-states like ``stopping.backends`` map to
+   ``state_stopping_backends`` (reference lib/pool.js:450 defines the
+   sub-state as a property of the parent state function).
+
+2. **Scoped resources.**  The ``S`` handle passed to a state-entry
+   function registers event listeners, timeouts and immediates that are
+   all torn down automatically when the state is exited
+   (lib/connection-fsm.js passim).
+
+3. **Synchronous chained transitions, deferred within entry.**  Signal
+   functions (``emit`` + listener + ``S.goto_state``) cause transitions
+   that complete synchronously from the caller's point of view — e.g.
+   ``fsm.set_unwanted()`` can leave the slot in state ``stopped`` by the
+   time it returns (relied on at lib/pool.js:624-631).  A transition
+   requested *while the entry function of the same FSM is still running*
+   is deferred until that entry function returns (this ordering is what
+   makes the claim double-handshake hand the user callback a slot that is
+   already ``busy``).
+
+4. **Asynchronous ``stateChanged`` emission.**  ``stateChanged`` events
+   are queued and delivered on the next loop turn, to the listeners
+   registered *at delivery time*.  Several reference code paths tolerate
+   — and depend on — observing a stale state on a queue while the real
+   transition event is still pending (lib/pool.js:937-946,
+   lib/connection-fsm.js:881-890, :1209-1216).
+"""
+
+from __future__ import annotations
+
+import asyncio
+from typing import Any, Callable, List, Optional, Sequence
+
+from .events import EventEmitter
+
+__all__ = ["FSM", "StateScope", "FSMError", "get_loop"]
+
+
+def get_loop(loop: Optional[asyncio.AbstractEventLoop] = None) -> asyncio.AbstractEventLoop:
+    if loop is not None:
+        return loop
+    try:
+        return asyncio.get_running_loop()
+    except RuntimeError:
+        return asyncio.get_event_loop()
+
+
+class FSMError(AssertionError):
+    """Invalid use of an FSM (bad transition, signal in wrong state...)."""
+
+
+class StateScope:
+    """The ``S`` handle given to state-entry functions.
+
+    Everything registered through the scope is disconnected when the FSM
+    leaves the state that created it.
+    """
+
+    __slots__ = ("_fsm", "_disposers", "_active")
+
+    def __init__(self, fsm: "FSM") -> None:
+        self._fsm = fsm
+        self._disposers: List[Callable[[], None]] = []
+        self._active = True
+
+    # -- queries ------------------------------------------------------
+    @property
+    def active(self) -> bool:
+        return self._active
+
+    # -- scoped registrations -----------------------------------------
+    def on(self, emitter: EventEmitter, event: str, cb: Callable) -> None:
+        if not self._active:
+            raise FSMError("S.on() used on exited state scope")
+        emitter.on(event, cb)
+        self._disposers.append(lambda: emitter.remove_listener(event, cb))
+
+    def timeout(self, ms: float, cb: Callable[[], None]) -> None:
+        """Run cb after `ms` milliseconds unless the state is exited first."""
+        if not self._active:
+            raise FSMError("S.timeout() used on exited state scope")
+        handle = self._fsm._loop.call_later(ms / 1000.0, self._guarded(cb))
+        self._disposers.append(handle.cancel)
+
+    def interval(self, ms: float, cb: Callable[[], None]) -> None:
+        if not self._active:
+            raise FSMError("S.interval() used on exited state scope")
+        state = {"h": None, "stop": False}
+
+        def tick() -> None:
+            if state["stop"] or not self._active:
+                return
+            cb()
+            if self._active and not state["stop"]:
+                state["h"] = self._fsm._loop.call_later(ms / 1000.0, tick)
+
+        state["h"] = self._fsm._loop.call_later(ms / 1000.0, tick)
+
+        def dispose() -> None:
+            state["stop"] = True
+            if state["h"] is not None:
+                state["h"].cancel()
+
+        self._disposers.append(dispose)
+
+    def immediate(self, cb: Callable[[], None]) -> None:
+        if not self._active:
+            raise FSMError("S.immediate() used on exited state scope")
+        handle = self._fsm._loop.call_soon(self._guarded(cb))
+        self._disposers.append(handle.cancel)
+
+    def callback(self, cb: Callable) -> Callable:
+        """Wrap cb so it becomes a no-op once the state has been exited."""
+
+        def wrapper(*args: Any) -> None:
+            if self._active:
+                cb(*args)
+
+        return wrapper
+
+    def _guarded(self, cb: Callable[[], None]) -> Callable[[], None]:
+        def wrapper() -> None:
+            if self._active:
+                cb()
+
+        return wrapper
+
+    # -- transitions ---------------------------------------------------
+    def valid_transitions(self, states: Sequence[str]) -> None:
+        self._fsm._fsm_valid = list(states)
+
+    def goto_state(self, state: str) -> None:
+        if not self._active:
+            # A handler belonging to an already-exited state fired during
+            # the same synchronous cascade; the transition it wanted is
+            # obsolete.
+            return
+        self._fsm.goto_state(state)
+
+    def goto_state_on(self, emitter: EventEmitter, event: str, state: str) -> None:
+        self.on(emitter, event, lambda *a: self.goto_state(state))
+
+    def goto_state_timeout(self, ms: float, state: str) -> None:
+        self.timeout(ms, lambda: self.goto_state(state))
+
+    # -- teardown ------------------------------------------------------
+    def _dispose(self) -> None:
+        self._active = False
+        disposers, self._disposers = self._disposers, []
+        for d in disposers:
+            d()
+
+
+class FSM(EventEmitter):
+    """Moore machine: outputs (entry actions) are a function of the state."""
+
+    __slots__ = (
+        "_loop",
+        "_fsm_state",
+        "_fsm_scope",
+        "_fsm_valid",
+        "_fsm_entering",
+        "_fsm_pending",
+        "_fsm_emit_queue",
+        "_fsm_emit_scheduled",
+        "_fsm_history",
+    )
+
+    #: ring-buffer length for state history (debugging aid; mooremachine
+    #: keeps history visible in core dumps)
+    HISTORY_LEN = 8
+
+    def __init__(self, initial_state: str,
+                 loop: Optional[asyncio.AbstractEventLoop] = None) -> None:
+        super().__init__()
+        self._loop = get_loop(loop)
+        self._fsm_state: Optional[str] = None
+        self._fsm_scope: Optional[StateScope] = None
+        self._fsm_valid: Optional[List[str]] = None
+        self._fsm_entering = False
+        self._fsm_pending: Optional[str] = None
+        self._fsm_emit_queue: List[str] = []
+        self._fsm_emit_scheduled = False
+        self._fsm_history: List[str] = []
+        self.goto_state(initial_state)
+
+    # -- introspection -------------------------------------------------
+    def get_state(self) -> str:
+        if self._fsm_state is None:
+            raise FSMError("FSM has no state yet")
+        return self._fsm_state
+
+    def is_in_state(self, state: str) -> bool:
+        cur = self._fsm_state
+        if cur is None:
+            return False
+        return cur == state or cur.startswith(state + ".")
+
+    def get_state_history(self) -> List[str]:
+        return list(self._fsm_history)
+
+    # -- transitions ---------------------------------------------------
+    def goto_state(self, state: str) -> None:
+        if self._fsm_valid is not None and state not in self._fsm_valid:
+            raise FSMError(
+                "%s: invalid transition %r -> %r (valid: %r)"
+                % (type(self).__name__, self._fsm_state, state, self._fsm_valid)
+            )
+        if self._fsm_entering:
+            # Requested while this FSM's entry function is still running:
+            # defer until it returns (see module docstring, point 3).
+            if self._fsm_pending is not None and self._fsm_pending != state:
+                raise FSMError(
+                    "%s: conflicting deferred transitions %r and %r from %r"
+                    % (type(self).__name__, self._fsm_pending, state,
+                       self._fsm_state)
+                )
+            self._fsm_pending = state
+            return
+        self._enter_loop(state)
+
+    def _enter_loop(self, state: str) -> None:
+        next_state: Optional[str] = state
+        while next_state is not None:
+            target = next_state
+            next_state = None
+            if self._fsm_scope is not None:
+                self._fsm_scope._dispose()
+            self._fsm_valid = None
+            self._fsm_state = target
+            self._fsm_history.append(target)
+            if len(self._fsm_history) > self.HISTORY_LEN:
+                del self._fsm_history[0]
+            scope = StateScope(self)
+            self._fsm_scope = scope
+            entry = getattr(self, "state_" + target.replace(".", "_"), None)
+            if entry is None:
+                raise FSMError(
+                    "%s has no state-entry function for %r"
+                    % (type(self).__name__, target)
+                )
+            self._fsm_entering = True
+            try:
+                entry(scope)
+            finally:
+                self._fsm_entering = False
+                pend, self._fsm_pending = self._fsm_pending, None
+            self._queue_state_changed(target)
+            if pend is not None:
+                if self._fsm_valid is not None and pend not in self._fsm_valid:
+                    raise FSMError(
+                        "%s: invalid transition %r -> %r (valid: %r)"
+                        % (type(self).__name__, target, pend, self._fsm_valid)
+                    )
+                next_state = pend
+
+    # -- async stateChanged delivery ------------------------------------
+    def _queue_state_changed(self, state: str) -> None:
+        self._fsm_emit_queue.append(state)
+        if not self._fsm_emit_scheduled:
+            self._fsm_emit_scheduled = True
+            self._loop.call_soon(self._flush_state_changed)
+
+    def _flush_state_changed(self) -> None:
+        self._fsm_emit_scheduled = False
+        while self._fsm_emit_queue:
+            st = self._fsm_emit_queue.pop(0)
+            self.emit("stateChanged", st)

@@ -1,0 +1,200 @@
+"""Test harness: virtual-time event loop and fake components.
+
+The reference's test kit (survey §4) drives multi-node behavior with
+in-process fakes: a fake resolver (test/pool.test.js:45-67), fake
+connections (test/pool.test.js:69-98) and a fake DNS client
+(test/dns.test.js:75-110).  This module provides the same kit for the
+rebuild, plus a deterministic virtual-clock event loop so backoff /
+CoDel / TTL logic can be tested in milliseconds of real time.
+"""
+
+from __future__ import annotations
+
+import asyncio
+import selectors
+from typing import Any, Callable, Dict, List, Optional
+
+from .events import EventEmitter
+
+__all__ = [
+    "VirtualLoop",
+    "advance",
+    "settle",
+    "run",
+    "DummyConnection",
+    "DummyResolver",
+]
+
+
+class VirtualLoop(asyncio.SelectorEventLoop):
+    """An asyncio event loop whose clock only moves when told to.
+
+    ``loop.time()`` returns virtual seconds.  Use ``await advance(loop, s)``
+    inside a coroutine running on the loop to fast-forward: all timers due
+    within the window fire in order, at their scheduled virtual times.
+    """
+
+    def __init__(self) -> None:
+        super().__init__(selectors.SelectSelector())
+        self._vtime = 1_000_000.0
+
+    def time(self) -> float:  # overrides BaseEventLoop.time
+        return self._vtime
+
+    def _vt_set(self, t: float) -> None:
+        if t < self._vtime:
+            raise ValueError("virtual time cannot go backwards")
+        self._vtime = t
+
+
+async def settle(loop: Optional[asyncio.AbstractEventLoop] = None,
+                 rounds: int = 1000) -> None:
+    """Let all ready (call_soon) callbacks run, including cascades."""
+    loop = loop or asyncio.get_running_loop()
+    for _ in range(rounds):
+        if not loop._ready:  # type: ignore[attr-defined]
+            return
+        await asyncio.sleep(0)
+    raise RuntimeError("event loop did not settle after %d rounds" % rounds)
+
+
+def _next_timer(loop: asyncio.AbstractEventLoop) -> Optional[float]:
+    times = [h._when for h in loop._scheduled  # type: ignore[attr-defined]
+             if not h._cancelled]
+    return min(times) if times else None
+
+
+async def advance(loop: asyncio.AbstractEventLoop, seconds: float) -> None:
+    """Fast-forward a VirtualLoop by `seconds`, firing due timers in order."""
+    if not isinstance(loop, VirtualLoop):
+        raise TypeError("advance() requires a VirtualLoop")
+    await settle(loop)
+    target = loop.time() + seconds
+    while True:
+        nxt = _next_timer(loop)
+        if nxt is None or nxt > target:
+            break
+        loop._vt_set(nxt)
+        await asyncio.sleep(0)  # let the due timers move to ready...
+        await settle(loop)      # ...and run, with any cascades
+    loop._vt_set(target)
+    await asyncio.sleep(0)
+    await settle(loop)
+
+
+def run(coro_fn: Callable[[asyncio.AbstractEventLoop], Any]) -> Any:
+    """Run async test body on a fresh VirtualLoop: run(lambda loop: body(loop))."""
+    loop = VirtualLoop()
+    try:
+        return loop.run_until_complete(coro_fn(loop))
+    finally:
+        loop.close()
+
+
+class DummyConnection(EventEmitter):
+    """Scripted fake of the user Connection interface (docs/api.adoc:580-645).
+
+    Tests construct pools with ``constructor=lambda backend:
+    DummyConnection(backend, registry)`` and then drive each instance by
+    hand: ``conn.connect()``, ``conn.emit('error', e)``, etc., exactly
+    like the reference's DummyConnection (test/pool.test.js:69-98).
+    """
+
+    def __init__(self, backend: Dict[str, Any],
+                 registry: Optional[List["DummyConnection"]] = None) -> None:
+        super().__init__()
+        self.backend = backend
+        self.connected = False
+        self.dead = False
+        self.ref_count = 0
+        self.seen_unwanted = False
+        if registry is not None:
+            registry.append(self)
+
+    def connect(self) -> None:
+        self.connected = True
+        self.emit("connect")
+
+    def destroy(self) -> None:
+        self.dead = True
+        self.connected = False
+
+    def set_unwanted(self) -> None:
+        self.seen_unwanted = True
+
+    def ref(self) -> None:
+        self.ref_count += 1
+
+    def unref(self) -> None:
+        self.ref_count -= 1
+
+
+def dummy_resolver_factory(registry: List["DummyResolver"]):
+    """Factory usable as a monkeypatch for ``cueball_amd.resolver.Resolver``
+    so a pool/set builds a DummyResolver instead of a DNS resolver — the
+    rebuild's equivalent of the reference's sinon stub
+    (test/pool.test.js:100-104).  Each created inner resolver is appended
+    to `registry`."""
+    from .resolver import ResolverFSM
+
+    def factory(options):
+        inner = DummyResolver()
+        registry.append(inner)
+        return ResolverFSM(inner, {"loop": options.get("loop")})
+
+    return factory
+
+
+class DummyResolver(EventEmitter):
+    """Bare fake of the Resolver interface (reference test/pool.test.js:45-67).
+
+    Drive topology from tests with ``resolver.add('b1', {"address": ...,
+    "port": ...})`` / ``resolver.remove('b1')``.
+    """
+
+    def __init__(self, backends: Optional[Dict[str, Dict[str, Any]]] = None) -> None:
+        super().__init__()
+        self.state = "stopped"
+        self.backends: Dict[str, Dict[str, Any]] = {}
+        self._initial = dict(backends or {})
+        self._last_error: Optional[BaseException] = None
+
+    # Resolver interface ------------------------------------------------
+    def start(self) -> None:
+        self.state = "running"
+        for k, b in self._initial.items():
+            self.add(k, b)
+
+    def stop(self) -> None:
+        self.state = "stopped"
+
+    def count(self) -> int:
+        return len(self.backends)
+
+    def list(self) -> Dict[str, Dict[str, Any]]:
+        return dict(self.backends)
+
+    def get_last_error(self) -> Optional[BaseException]:
+        return self._last_error
+
+    def is_in_state(self, state: str) -> bool:
+        return self.state == state
+
+    def get_state(self) -> str:
+        return self.state
+
+    # test-side controls ------------------------------------------------
+    def add(self, key: str, backend: Dict[str, Any]) -> None:
+        backend = dict(backend)
+        backend["key"] = key
+        self.backends[key] = backend
+        self.emit("added", key, backend)
+
+    def remove(self, key: str) -> None:
+        self.backends.pop(key, None)
+        self.emit("removed", key)
+
+    def fail(self, err: BaseException) -> None:
+        self._last_error = err
+        self.state = "failed"
+        self.emit("stateChanged", "failed")

@@ -1,0 +1,601 @@
+"""ConnectionPool behavior tests.
+
+Port of the reference's pool scenarios (test/pool.test.js) onto the
+virtual-clock harness: fake resolver + scripted DummyConnections give
+deterministic control of every connect/fail/close ordering, including
+the race regressions.
+"""
+
+import math
+
+import pytest
+
+from cueball_amd.errors import (ClaimTimeoutError, NoBackendsError,
+                                PoolFailedError, PoolStoppingError)
+from cueball_amd.pool import ConnectionPool
+from cueball_amd.resolver import ResolverFSM
+from cueball_amd.testing import (DummyConnection, DummyResolver, advance,
+                                 settle)
+from conftest import run_vt
+
+RECOVERY = {"default": {"timeout": 500, "retries": 1, "delay": 0}}
+
+
+class Ctx:
+    """One test's kit: pool + injected resolver + connection registry."""
+
+    def __init__(self, loop, spares=2, maximum=4, recovery=None, **opts):
+        self.loop = loop
+        self.connections = []
+        self.resolver = DummyResolver()
+        self.resolver_fsm = ResolverFSM(self.resolver, {"loop": loop})
+
+        def constructor(backend):
+            c = DummyConnection(backend)
+            c.backend = backend.get("key")
+            self.connections.append(c)
+            orig_destroy = c.destroy
+
+            def destroy():
+                if c in self.connections:
+                    self.connections.remove(c)
+                orig_destroy()
+
+            c.destroy = destroy
+            return c
+
+        pool_opts = {
+            "domain": "foobar",
+            "constructor": constructor,
+            "recovery": recovery or RECOVERY,
+            "spares": spares,
+            "maximum": maximum,
+            "resolver": self.resolver_fsm,
+            "loop": loop,
+        }
+        pool_opts.update(opts)
+        self.pool = ConnectionPool(pool_opts)
+        # like the reference's stubbed-in resolver, the pool owns it:
+        # start it ourselves since we passed a custom resolver
+        self.resolver_fsm.start()
+
+    def counts(self):
+        out = {}
+        for c in self.connections:
+            out[c.backend] = out.get(c.backend, 0) + 1
+        return out
+
+    def by_backend(self, key):
+        return [c for c in self.connections if c.backend == key]
+
+    def add_backend(self, key):
+        self.resolver.add(key, {})
+
+    def claim(self, opts=None):
+        """Returns (handle_box, result_box): result_box gets (err, hdl,
+        conn) when the callback fires."""
+        box = {}
+
+        def cb(err, hdl=None, conn=None):
+            box["err"] = err
+            box["hdl"] = hdl
+            box["conn"] = conn
+
+        ret = self.pool.claim(opts or {}, cb)
+        return ret, box
+
+
+def test_empty_pool_error_on_empty_and_timeout():
+    async def body(loop):
+        ctx = Ctx(loop, spares=2, maximum=4)
+        await settle(loop)
+        assert len(ctx.connections) == 0
+
+        _, box = ctx.claim({"errorOnEmpty": True})
+        await settle(loop)
+        assert isinstance(box["err"], NoBackendsError)
+
+        _, box2 = ctx.claim({"timeout": 100})
+        await advance(loop, 0.2)
+        assert isinstance(box2["err"], ClaimTimeoutError)
+
+    run_vt(lambda loop: body(loop))
+
+
+def test_pool_with_one_backend():
+    async def body(loop):
+        ctx = Ctx(loop, spares=2, maximum=2)
+        ctx.add_backend("b1")
+        await settle(loop)
+        assert len(ctx.connections) == 2
+        assert all(c.backend == "b1" for c in ctx.connections)
+
+        # connections haven't connected yet: claims time out
+        _, box = ctx.claim({"timeout": 100})
+        await advance(loop, 0.2)
+        assert isinstance(box["err"], ClaimTimeoutError)
+
+        for c in list(ctx.connections):
+            c.connect()
+        await settle(loop)
+
+        _, box2 = ctx.claim({"timeout": 100})
+        await settle(loop)
+        assert box2["err"] is None
+        assert box2["conn"] in ctx.connections
+
+        _, box3 = ctx.claim({"timeout": 100})
+        await settle(loop)
+        assert box3["err"] is None
+        assert box3["conn"] is not box2["conn"]
+
+        # all busy now: next claim times out
+        _, box4 = ctx.claim({"timeout": 100})
+        await advance(loop, 0.2)
+        assert isinstance(box4["err"], ClaimTimeoutError)
+
+    run_vt(lambda loop: body(loop))
+
+
+def test_async_claim_expands_to_max():
+    async def body(loop):
+        ctx = Ctx(loop, spares=0, maximum=2)
+        ctx.add_backend("b1")
+        ctx.add_backend("b2")
+        await settle(loop)
+        assert len(ctx.connections) == 0  # spares=0
+
+        _, box1 = ctx.claim()
+        await settle(loop)
+        # claim queued -> rebalance creates a connection
+        assert len(ctx.connections) == 1
+        ctx.connections[0].connect()
+        await settle(loop)
+        assert box1["err"] is None
+        b1 = box1["conn"].backend
+
+        _, box2 = ctx.claim()
+        await settle(loop)
+        assert len(ctx.connections) == 2
+        ctx.connections[1].connect()
+        await settle(loop)
+        assert box2["err"] is None
+        b2 = box2["conn"].backend
+        assert {b1, b2} == {"b1", "b2"}
+
+        # at max: a further claim times out
+        _, box3 = ctx.claim({"timeout": 100})
+        await advance(loop, 0.2)
+        assert isinstance(box3["err"], ClaimTimeoutError)
+
+    run_vt(lambda loop: body(loop))
+
+
+def test_spares_evenly_balanced():
+    async def body(loop):
+        ctx = Ctx(loop, spares=4, maximum=4)
+        ctx.add_backend("b1")
+        ctx.add_backend("b2")
+        await settle(loop)
+        for c in list(ctx.connections):
+            c.connect()
+        await settle(loop)
+        assert len(ctx.connections) == 4
+        bs = sorted(c.backend for c in ctx.connections)
+        assert bs == ["b1", "b1", "b2", "b2"]
+
+        ctx.add_backend("b3")
+        ctx.add_backend("b4")
+        await advance(loop, 0.1)
+        for c in list(ctx.connections):
+            if not c.connected:
+                c.connect()
+        await advance(loop, 0.1)
+        # rebalance needs a couple of cycles to drain/add
+        for c in list(ctx.connections):
+            if not c.connected:
+                c.connect()
+        await advance(loop, 0.1)
+        assert len(ctx.connections) == 4
+        bs2 = sorted(c.backend for c in ctx.connections)
+        assert bs2 == ["b1", "b2", "b3", "b4"]
+
+    run_vt(lambda loop: body(loop))
+
+
+def test_error_while_claimed():
+    async def body(loop):
+        ctx = Ctx(loop, spares=1, maximum=1)
+        ctx.add_backend("b1")
+        await settle(loop)
+        assert len(ctx.connections) == 1
+        ctx.connections[0].connect()
+        await settle(loop)
+
+        _, box = ctx.claim()
+        await settle(loop)
+        conn = box["conn"]
+        assert conn is not None
+        seen = []
+        conn.once("error", seen.append)
+        conn.emit("error", RuntimeError("testing"))
+        assert len(seen) == 1
+        box["hdl"].release()
+        await advance(loop, 0.5)
+        assert conn.dead
+        # the pool replaced it with a fresh connection
+        assert len(ctx.connections) == 1
+        assert ctx.connections[0] is not conn
+        ctx.connections[0].connect()
+        await settle(loop)
+        assert ctx.pool.is_in_state("running")
+
+    run_vt(lambda loop: body(loop))
+
+
+def test_error_while_claimed_no_listener_raises():
+    """If the claimer registered no 'error' listener, the error is
+    re-raised loudly (lib/connection-fsm.js:697-706)."""
+    async def body(loop):
+        ctx = Ctx(loop, spares=1, maximum=1)
+        ctx.add_backend("b1")
+        await settle(loop)
+        ctx.connections[0].connect()
+        await settle(loop)
+        _, box = ctx.claim()
+        await settle(loop)
+        conn = box["conn"]
+        with pytest.raises(RuntimeError):
+            conn.emit("error", RuntimeError("unhandled"))
+
+    run_vt(lambda loop: body(loop))
+
+
+def test_close_while_idle():
+    async def body(loop):
+        ctx = Ctx(loop, spares=1, maximum=1)
+        ctx.add_backend("b1")
+        await settle(loop)
+        conn = ctx.connections[0]
+        conn.connect()
+        await advance(loop, 0.1)
+
+        conn.emit("close")
+        await settle(loop)
+        assert conn.dead
+        # replacement created straight away, no backoff on clean close
+        assert len(ctx.connections) == 1
+        assert ctx.connections[0] is not conn
+        assert not ctx.connections[0].dead
+        smgr_hist = conn.sm_fsm.get_state_history()
+        assert "backoff" not in smgr_hist
+        ctx.connections[0].connect()
+        await settle(loop)
+
+        states = []
+        ctx.pool.on("stateChanged", states.append)
+        ctx.pool.stop()
+        await advance(loop, 2.0)
+        assert ctx.pool.is_in_state("stopped")
+
+    run_vt(lambda loop: body(loop))
+
+
+def test_removing_a_backend():
+    async def body(loop):
+        ctx = Ctx(loop, spares=2, maximum=3)
+        ctx.add_backend("b1")
+        ctx.add_backend("b2")
+        await settle(loop)
+        assert ctx.counts() == {"b1": 1, "b2": 1}
+        ctx.by_backend("b1")[0].connect()
+        # kill b2 until it's declared dead (retries=1 -> immediate)
+        ctx.by_backend("b2")[0].emit("error", RuntimeError())
+        await advance(loop, 0.4)
+
+        assert list(ctx.pool.p_dead.keys()) == [
+            k for k in ctx.pool.p_dead]
+        assert len(ctx.pool.p_dead) == 1
+        # monitor on b2 + replacement on b1
+        assert ctx.counts() == {"b1": 2, "b2": 1}
+
+        ctx.by_backend("b1")[1].connect()
+        conn = ctx.by_backend("b2")[0]
+
+        ctx.resolver.remove("b2")
+        await advance(loop, 0.8)
+        # monitor may still be retrying: kill it again if present
+        if ctx.counts().get("b2", 0) > 0:
+            ctx.by_backend("b2")[0].emit("error", RuntimeError())
+        await advance(loop, 1.0)
+        assert conn.dead
+        assert ctx.counts() == {"b1": 2}
+
+        ctx.pool.stop()
+        await advance(loop, 2.0)
+        assert ctx.pool.is_in_state("stopped")
+
+    run_vt(lambda loop: body(loop))
+
+
+def test_pool_failure_and_recovery():
+    async def body(loop):
+        recovery = {"default": {"timeout": 500, "retries": 2, "delay": 0}}
+        ctx = Ctx(loop, spares=2, maximum=2, recovery=recovery)
+        ctx.add_backend("b1")
+        await settle(loop)
+        assert ctx.counts() == {"b1": 2}
+
+        conns = list(ctx.connections)
+        conns[0].connect()
+        conns[0].emit("error", RuntimeError())
+        conns[1].connect()
+        conns[1].emit("error", RuntimeError())
+        await advance(loop, 0.1)
+        assert ctx.pool.is_in_state("running")
+        assert len(ctx.connections) == 2
+
+        conns = list(ctx.connections)
+        conns[1].connect()
+        conns[1].emit("error", RuntimeError())
+        conns[0].connect()
+        await advance(loop, 0.1)
+        assert ctx.pool.is_in_state("running")
+        assert len(ctx.connections) == 2
+
+        conns = list(ctx.connections)
+        conns[0].emit("error", RuntimeError("test"))
+        conns[1].emit("error", RuntimeError("test"))
+
+        _, box = ctx.claim()
+        await advance(loop, 0.1)
+        assert ctx.pool.is_in_state("failed")
+        assert isinstance(box["err"], PoolFailedError)
+        assert ctx.pool.get_last_error() is not None
+
+        # only the monitor remains
+        assert ctx.counts() == {"b1": 1}
+        ctx.connections[0].connect()
+        await advance(loop, 0.1)
+        assert ctx.pool.is_in_state("running")
+
+        ctx.pool.stop()
+        await advance(loop, 2.0)
+        assert ctx.pool.is_in_state("stopped")
+
+    run_vt(lambda loop: body(loop))
+
+
+def test_pool_failure_retry_race():
+    """Mixed connect/error orderings across retry rounds must not
+    wedge the pool or lose slots (test/pool.test.js:540)."""
+    async def body(loop):
+        recovery = {"default": {"timeout": 500, "retries": 2, "delay": 0}}
+        ctx = Ctx(loop, spares=2, maximum=2, recovery=recovery)
+        ctx.add_backend("b1")
+        await settle(loop)
+        assert ctx.counts() == {"b1": 2}
+
+        conns = list(ctx.connections)
+        conns[0].connect()
+        conns[0].emit("error", RuntimeError("test"))
+        conns[1].connect()
+        conns[1].emit("error", RuntimeError("test"))
+        await advance(loop, 0.1)
+        assert ctx.pool.is_in_state("running")
+        assert len(ctx.connections) == 2
+
+        conns = list(ctx.connections)
+        conns[1].connect()
+        conns[1].emit("error", RuntimeError("test"))
+        conns[0].connect()
+        conns[0].emit("error", RuntimeError("test"))
+        await advance(loop, 0.1)
+        assert ctx.pool.is_in_state("running")
+        assert ctx.pool.get_last_error() is None
+
+        assert len(ctx.connections) == 2
+        conns = list(ctx.connections)
+        conns[1].emit("error", RuntimeError("test2"))
+        conns[0].connect()
+        await advance(loop, 0.1)
+        assert ctx.pool.is_in_state("running")
+        assert ctx.counts() == {"b1": 2}
+
+        ctx.pool.stop()
+        await advance(loop, 2.0)
+        assert ctx.pool.is_in_state("stopped")
+
+    run_vt(lambda loop: body(loop))
+
+
+def test_ping_checker():
+    async def body(loop):
+        checked = []
+
+        def do_check(hdl, conn):
+            conn.checked = True
+            checked.append(conn)
+            hdl.release()
+
+        ctx = Ctx(loop, spares=2, maximum=2, checkTimeout=100,
+                  checker=do_check)
+        for c in ctx.connections:
+            c.checked = False
+        ctx.add_backend("b1")
+        await settle(loop)
+        assert len(ctx.connections) == 2
+        for c in ctx.connections:
+            c.checked = False
+            c.connect()
+        await settle(loop)
+
+        _, box = ctx.claim()
+        await settle(loop)
+        conn = box["conn"]
+        assert getattr(conn, "checked", False) is False
+
+        await advance(loop, 1.0)
+        # the idle conn got pinged; the busy one didn't
+        cs = sorted(bool(getattr(c, "checked", False))
+                    for c in ctx.connections)
+        assert cs == [False, True]
+        box["hdl"].release()
+
+        ctx.pool.stop()
+        await settle(loop)
+
+    run_vt(lambda loop: body(loop))
+
+
+def test_ping_checker_does_not_expand_pool():
+    async def body(loop):
+        def do_check(hdl, conn):
+            conn.checked = True
+            hdl.release()
+
+        ctx = Ctx(loop, spares=2, maximum=10, checkTimeout=100,
+                  checker=do_check)
+        ctx.add_backend("b1")
+        await settle(loop)
+        assert len(ctx.connections) == 2
+        for c in ctx.connections:
+            c.checked = False
+            c.connect()
+        await settle(loop)
+
+        await advance(loop, 0.3)
+        assert all(getattr(c, "checked", False) for c in ctx.connections)
+        # pings ride the initq: the pool must NOT have grown
+        assert len(ctx.connections) == 2
+
+        ctx.pool.stop()
+        await settle(loop)
+
+    run_vt(lambda loop: body(loop))
+
+
+def test_claim_cancellation():
+    async def body(loop):
+        ctx = Ctx(loop, spares=2, maximum=2)
+        ctx.add_backend("b1")
+        await settle(loop)
+        assert len(ctx.connections) == 2
+
+        _, box = ctx.claim({"timeout": 100})
+        await advance(loop, 0.2)
+        assert isinstance(box["err"], ClaimTimeoutError)
+
+        for c in ctx.connections:
+            c.connect()
+        await settle(loop)
+
+        fired = []
+        handle = ctx.pool.claim({"timeout": 100},
+                                lambda *a: fired.append(a))
+        handle.cancel()
+        await advance(loop, 0.15)
+        assert fired == []  # callback never runs after cancel()
+
+        ctx.pool.stop()
+        await advance(loop, 2.0)
+        assert ctx.pool.is_in_state("stopped")
+
+    run_vt(lambda loop: body(loop))
+
+
+def test_claim_on_stopped_pool():
+    async def body(loop):
+        ctx = Ctx(loop, spares=1, maximum=1)
+        ctx.add_backend("b1")
+        await settle(loop)
+        ctx.pool.stop()
+        await settle(loop)
+
+        _, box = ctx.claim()
+        await settle(loop)
+        assert isinstance(box["err"], PoolStoppingError)
+
+    run_vt(lambda loop: body(loop))
+
+
+def test_get_stats():
+    async def body(loop):
+        ctx = Ctx(loop, spares=2, maximum=4)
+        ctx.add_backend("b1")
+        await settle(loop)
+        for c in ctx.connections:
+            c.connect()
+        await settle(loop)
+
+        stats = ctx.pool.get_stats()
+        assert stats["totalConnections"] == 2
+        assert stats["idleConnections"] == 2
+        assert stats["pendingConnections"] == 0
+        assert stats["waiterCount"] == 0
+
+        _, box = ctx.claim()
+        await settle(loop)
+        stats = ctx.pool.get_stats()
+        assert stats["idleConnections"] == 1
+        assert stats["counters"]["claim"] == 1
+
+        box["hdl"].release()
+        await settle(loop)
+        stats = ctx.pool.get_stats()
+        assert stats["idleConnections"] == 2
+
+    run_vt(lambda loop: body(loop))
+
+
+def test_release_twice_raises():
+    async def body(loop):
+        ctx = Ctx(loop, spares=1, maximum=1)
+        ctx.add_backend("b1")
+        await settle(loop)
+        ctx.connections[0].connect()
+        await settle(loop)
+        _, box = ctx.claim()
+        await settle(loop)
+        box["hdl"].release()
+        with pytest.raises(Exception) as ei:
+            box["hdl"].release()
+        assert "released by" in str(ei.value)
+
+    run_vt(lambda loop: body(loop))
+
+
+def test_claim_handle_misuse_traps():
+    async def body(loop):
+        from cueball_amd.errors import ClaimHandleMisusedError
+        ctx = Ctx(loop, spares=1, maximum=1)
+        ctx.add_backend("b1")
+        await settle(loop)
+        ctx.connections[0].connect()
+        await settle(loop)
+        _, box = ctx.claim()
+        await settle(loop)
+        hdl = box["hdl"]
+        with pytest.raises(ClaimHandleMisusedError):
+            hdl.readable
+        with pytest.raises(ClaimHandleMisusedError):
+            hdl.writable
+        with pytest.raises(ClaimHandleMisusedError):
+            hdl.on("close", lambda: None)
+        box["hdl"].release()
+
+    run_vt(lambda loop: body(loop))
+
+
+def test_claim_async_sugar():
+    async def body(loop):
+        ctx = Ctx(loop, spares=1, maximum=1)
+        ctx.add_backend("b1")
+        await settle(loop)
+        ctx.connections[0].connect()
+        await settle(loop)
+        hdl, conn = await ctx.pool.claim_async()
+        assert conn is ctx.connections[0]
+        hdl.release()
+
+    run_vt(lambda loop: body(loop))
