@@ -286,10 +286,14 @@ class DNSResolverFSM(FSM):
         self.r_bootstrap: Optional["DNSResolverFSM"] = None
         self.r_bootstrap_res: Dict[str, Dict[str, Any]] = {}
 
-        nsclient = DNSResolverFSM.global_ns_clients.get(self.r_maxres)
+        # tests inject a scripted client here, like the reference's sinon
+        # stub of mname-client (test/dns.test.js:75-110)
+        nsclient = options.get("_nsclient")
         if nsclient is None:
-            nsclient = DnsClient(concurrency=self.r_maxres)
-            DNSResolverFSM.global_ns_clients[self.r_maxres] = nsclient
+            nsclient = DNSResolverFSM.global_ns_clients.get(self.r_maxres)
+            if nsclient is None:
+                nsclient = DnsClient(concurrency=self.r_maxres)
+                DNSResolverFSM.global_ns_clients[self.r_maxres] = nsclient
         self.r_nsclient = nsclient
 
         self.r_stopping = False
@@ -339,7 +343,11 @@ class DNSResolverFSM(FSM):
 
     def state_check_ns(self, S: StateScope) -> None:
         if self.r_resolvers:
-            not_ip = [r for r in self.r_resolvers if not _is_ip(r)]
+            # resolvers may use "ip@port" to target a non-53 port
+            # (tests/mock servers); only a bare non-IP string means
+            # Dynamic Resolver mode
+            not_ip = [r for r in self.r_resolvers
+                      if not _is_ip(r.split("@", 1)[0])]
             if not not_ip:
                 S.goto_state("srv")
                 return
@@ -819,7 +827,7 @@ class DNSResolverFSM(FSM):
             opts["errorThreshold"] = min(self.r_maxres,
                                          len(self.r_resolvers)) or 1
 
-        em = EventEmitter()
+        em = _DnsRequest()
 
         def on_lookup(err: Optional[BaseException], msg: Any) -> None:
             # Vote on the most common rcode across a MultiError
@@ -917,6 +925,13 @@ class DNSResolverFSM(FSM):
 
         em.send = send  # type: ignore[attr-defined]
         return em
+
+
+class _DnsRequest(EventEmitter):
+    """One in-flight DNS question: emits 'answers'(ans, ttl) or
+    'error'(err) after .send()."""
+
+    send: Any = None
 
 
 def _verror(cause: Optional[BaseException], msg: str) -> BaseException:

@@ -12,8 +12,9 @@ from __future__ import annotations
 
 import asyncio
 import selectors
-from typing import Any, Callable, Dict, List, Optional
+from typing import Any, Callable, Dict, List, Optional, Tuple
 
+from . import dns_wire
 from .events import EventEmitter
 
 __all__ = [
@@ -23,6 +24,7 @@ __all__ = [
     "run",
     "DummyConnection",
     "DummyResolver",
+    "MockDnsServer",
 ]
 
 
@@ -198,3 +200,95 @@ class DummyResolver(EventEmitter):
         self._last_error = err
         self.state = "failed"
         self.emit("stateChanged", "failed")
+
+
+class MockDnsServer:
+    """A real UDP DNS server on 127.0.0.1 serving a scripted zone.
+
+    Used by the dns_client integration tests and the DNS-SRV benchmark
+    config (BASELINE.json config #2).  The zone maps (name, type) to a
+    list of record dicts in dns_wire format, e.g.::
+
+        srv.add_srv("_http._tcp.svc", "b1.svc", 8080, ttl=30)
+        srv.add_a("b1.svc", "127.0.0.1", ttl=30)
+    """
+
+    def __init__(self) -> None:
+        self.zone: Dict[Tuple[str, str], List[Dict[str, Any]]] = {}
+        self.queries: List[Tuple[str, str]] = []
+        self.port: Optional[int] = None
+        self.drop_next = 0           # drop this many queries (timeouts)
+        self.rcode_override: Optional[str] = None
+        self._transport = None
+
+    # -- zone building --------------------------------------------------
+    def add_srv(self, name: str, target: str, port: int,
+                ttl: int = 60) -> None:
+        self.zone.setdefault((name.lower(), "SRV"), []).append({
+            "type": "SRV", "name": name, "ttl": ttl, "priority": 0,
+            "weight": 10, "port": port, "target": target})
+
+    def add_a(self, name: str, address: str, ttl: int = 60) -> None:
+        self.zone.setdefault((name.lower(), "A"), []).append({
+            "type": "A", "name": name, "ttl": ttl, "target": address})
+
+    def add_aaaa(self, name: str, address: str, ttl: int = 60) -> None:
+        self.zone.setdefault((name.lower(), "AAAA"), []).append({
+            "type": "AAAA", "name": name, "ttl": ttl, "target": address})
+
+    def clear(self) -> None:
+        self.zone.clear()
+
+    # -- lifecycle -------------------------------------------------------
+    async def start(self, port: int = 0) -> int:
+        loop = asyncio.get_running_loop()
+        server = self
+
+        class Proto(asyncio.DatagramProtocol):
+            def connection_made(self, transport):
+                self.transport = transport
+
+            def datagram_received(self, data, addr):
+                resp = server._handle(data)
+                if resp is not None:
+                    self.transport.sendto(resp, addr)
+
+        self._transport, _ = await loop.create_datagram_endpoint(
+            Proto, local_addr=("127.0.0.1", port))
+        self.port = self._transport.get_extra_info("sockname")[1]
+        return self.port
+
+    def stop(self) -> None:
+        if self._transport is not None:
+            self._transport.close()
+            self._transport = None
+
+    @property
+    def resolver_address(self) -> str:
+        return "127.0.0.1@%d" % self.port
+
+    # -- request handling -------------------------------------------------
+    def _handle(self, data: bytes) -> Optional[bytes]:
+        try:
+            q = dns_wire.decode_message(data)
+        except ValueError:
+            return None
+        if not q.question:
+            return None
+        if self.drop_next > 0:
+            self.drop_next -= 1
+            return None
+        question = q.question[0]
+        name, rtype = question["name"], question["type"]
+        self.queries.append((name, rtype))
+        if self.rcode_override is not None:
+            return dns_wire.encode_response(q.id, question,
+                                            rcode=self.rcode_override)
+        answers = self.zone.get((name.lower(), rtype))
+        if answers is None:
+            if any(k[0] == name.lower() for k in self.zone):
+                # NODATA: name exists, no records of this type
+                return dns_wire.encode_response(q.id, question, authority=[{
+                    "type": "SOA", "name": name, "ttl": 60}])
+            return dns_wire.encode_response(q.id, question, rcode="NXDOMAIN")
+        return dns_wire.encode_response(q.id, question, answers=answers)
