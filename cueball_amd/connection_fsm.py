@@ -304,10 +304,12 @@ class ClaimHandle(FSM):
         self.ch_pool = options["pool"]
         throw_error = options.get("throwError")
         self.ch_throw_error = True if throw_error is None else bool(throw_error)
-        self.ch_claim_stack: List[str] = list(options["claimStack"])
+        self.ch_claim_stack: List[str] = options["claimStack"]
         self.ch_callback: Callable = options["callback"]
         log: CueballLogger = options["log"]
-        self.ch_log = log.child(component="ClaimHandle")
+        # hot path: the pool passes a pre-made child logger
+        self.ch_log = log if options.get("_logReady") \
+            else log.child(component="ClaimHandle")
 
         self.ch_slot: Optional["ConnectionSlotFSM"] = None
         self.ch_release_stack: Optional[List[str]] = None
@@ -344,45 +346,66 @@ class ClaimHandle(FSM):
         self.ch_do_release_leak_check = False
 
     # -- signal functions ----------------------------------------------
+    #
+    # These are direct state-dispatch rather than emit + scoped-listener
+    # pairs: the signals are internal to this FSM (nothing ever listens
+    # for them externally), and the claim path walks four handle states
+    # per claim — the listener churn was the hot spot.  goto_state still
+    # defers transitions requested during an entry function, so the
+    # observable ordering is identical to the listener formulation.
+
     def try_(self, slot: "ConnectionSlotFSM") -> None:
         """The pool offers `slot` to this handle (must be idle)."""
-        if not self.is_in_state("waiting"):
+        if self._fsm_state != "waiting":
             raise FSMError('ClaimHandle.try_ only in "waiting" (is in "%s")'
                            % self.get_state())
         if not slot.is_in_state("idle"):
             raise FSMError('ClaimHandle.try_ needs an idle slot (is in "%s")'
                            % slot.get_state())
         self.ch_slot = slot
-        self.emit("tryAsserted")
+        self.goto_state("claiming")
 
     def accept(self, connection: Any) -> None:
-        if not self.is_in_state("claiming"):
+        if self._fsm_state != "claiming":
             raise FSMError("accept only in claiming")
         self.ch_connection = connection
-        self.emit("accepted")
+        self.goto_state("claimed")
 
     def reject(self) -> None:
-        if not self.is_in_state("claiming"):
+        if self._fsm_state != "claiming":
             raise FSMError("reject only in claiming")
-        self.emit("rejected")
+        if self.ch_cancelled:
+            self.goto_state("cancelled")
+        else:
+            self.goto_state("waiting")
 
     def cancel(self) -> None:
-        if self.is_in_state("claimed"):
+        if self._fsm_state == "claimed":
             self.release()
         else:
             self.ch_cancelled = True
-            self.emit("cancelled")
+            # in "claiming" the cancellation is applied on reject/accept;
+            # in "waiting" it takes effect now (lib/connection-fsm.js:580)
+            if self._fsm_state == "waiting":
+                self.goto_state("cancelled")
 
     def timeout(self) -> None:
-        if not self.is_in_state("waiting"):
+        if self._fsm_state != "waiting":
             raise FSMError("timeout only in waiting")
-        self.emit("timeout")
+        self._on_claim_timeout()
+
+    def _on_claim_timeout(self) -> None:
+        self.ch_last_error = mod_errors.ClaimTimeoutError(self.ch_pool)
+        self.ch_pool._incr_counter("claim-timeout")
+        self.goto_state("failed")
 
     def fail(self, err: BaseException) -> None:
-        self.emit("error", err)
+        if self._fsm_state == "waiting":
+            self.ch_last_error = err
+            self.goto_state("failed")
 
-    def _relinquish(self, event: str) -> None:
-        if not self.is_in_state("claimed"):
+    def _relinquish(self, state: str) -> None:
+        if self._fsm_state != "claimed":
             if self.is_in_state("released") or self.is_in_state("closed"):
                 stack = self.ch_release_stack or ["?", "?", "?"]
                 by = stack[2] if len(stack) > 2 else stack[-1]
@@ -393,56 +416,27 @@ class ClaimHandle(FSM):
                 'ClaimHandle.release() called while in state "%s"'
                 % self.get_state())
         self.ch_release_stack = mod_utils.maybe_capture_stack_trace()
-        self.emit(event)
+        self.goto_state(state)
 
     def release(self) -> None:
-        self._relinquish("releaseAsserted")
+        self._relinquish("released")
 
     def close(self) -> None:
-        self._relinquish("closeAsserted")
+        self._relinquish("closed")
 
     # -- states ----------------------------------------------------------
     def state_waiting(self, S: StateScope) -> None:
-        S.valid_transitions(["claiming", "cancelled", "failed"])
+        S.valid_transitions(("claiming", "cancelled", "failed"))
         self.ch_slot = None
-
-        S.on(self, "tryAsserted", lambda: S.goto_state("claiming"))
-
-        def on_timeout() -> None:
-            self.ch_last_error = mod_errors.ClaimTimeoutError(self.ch_pool)
-            self.ch_pool._incr_counter("claim-timeout")
-            S.goto_state("failed")
-
         if math.isfinite(self.ch_claim_timeout):
-            S.timeout(self.ch_claim_timeout, on_timeout)
-        S.on(self, "timeout", on_timeout)
-
-        def on_error(err: BaseException) -> None:
-            self.ch_last_error = err
-            S.goto_state("failed")
-
-        S.on(self, "error", on_error)
-        S.on(self, "cancelled", lambda: S.goto_state("cancelled"))
+            S.timeout(self.ch_claim_timeout, self._on_claim_timeout)
 
     def state_claiming(self, S: StateScope) -> None:
-        S.valid_transitions(["claimed", "waiting", "cancelled"])
-
-        S.on(self, "accepted", lambda: S.goto_state("claimed"))
-
-        def on_rejected() -> None:
-            if self.ch_cancelled:
-                S.goto_state("cancelled")
-            else:
-                S.goto_state("waiting")
-
-        S.on(self, "rejected", on_rejected)
+        S.valid_transitions(("claimed", "waiting", "cancelled"))
         self.ch_slot.claim(self)
 
     def state_claimed(self, S: StateScope) -> None:
-        S.valid_transitions(["released", "closed"])
-
-        S.on(self, "releaseAsserted", lambda: S.goto_state("released"))
-        S.on(self, "closeAsserted", lambda: S.goto_state("closed"))
+        S.valid_transitions(("released", "closed"))
 
         if self.ch_cancelled:
             S.goto_state("released")
@@ -500,10 +494,18 @@ class ClaimHandle(FSM):
 def count_listeners(emitter: Any, event: str) -> int:
     """Count user-registered listeners, ignoring cueball's own internal
     handlers (lib/connection-fsm.js:786-808)."""
-    if not hasattr(emitter, "listeners"):
+    # fast path for our own EventEmitter: read the list in place
+    ev = getattr(emitter, "_events", None)
+    if ev is not None:
+        ls = ev.get(event)
+    elif hasattr(emitter, "listeners"):
+        ls = emitter.listeners(event)
+    else:
+        return 0
+    if not ls:
         return 0
     n = 0
-    for h in emitter.listeners(event):
+    for h in ls:
         if not callable(h):
             continue
         target = getattr(h, "listener", h)
@@ -560,12 +562,13 @@ class ConnectionSlotFSM(FSM):
         self.emit("startAsserted")
 
     def claim(self, handle: ClaimHandle) -> None:
-        if not self.is_in_state("idle"):
+        # direct dispatch (hot path; see ClaimHandle signal comment)
+        if self._fsm_state != "idle":
             raise FSMError("claim only in idle")
         if self.csf_handle is not None:
             raise FSMError("slot already has a handle")
         self.csf_handle = handle
-        self.emit("claimAsserted")
+        self.goto_state("busy")
 
     def make_child_logger(self, **fields: Any) -> CueballLogger:
         return self.csf_log.child(**fields)
@@ -677,7 +680,6 @@ class ConnectionSlotFSM(FSM):
                     'Unhandled smgr state transition: connected => "%s"' % st)
 
         S.on(smgr, "stateChanged", on_smgr_state)
-        S.on(self, "claimAsserted", lambda: S.goto_state("busy"))
 
         if self.csf_check_timeout is not None and \
                 self.csf_checker is not None:

@@ -76,6 +76,11 @@ class EventEmitter:
         ls = self._events.get(event)
         if not ls:
             return False
+        if len(ls) == 1:
+            # copy-free fast path; snapshot semantics still hold (a
+            # listener added during this call is not invoked)
+            ls[0](*args)
+            return True
         for listener in tuple(ls):
             listener(*args)
         return True

@@ -33,12 +33,17 @@ depends on (survey §7 "hard parts" #1):
    — and depend on — observing a stale state on a queue while the real
    transition event is still pending (lib/pool.js:937-946,
    lib/connection-fsm.js:881-890, :1209-1216).
+
+The runtime sits on the claim hot path (a claim/release cycle walks
+6+ state transitions across three FSMs), so it is written for low
+allocation: per-class state-entry method caches, tuple-based scope
+disposal, and a copy-free single-listener emit fast path.
 """
 
 from __future__ import annotations
 
 import asyncio
-from typing import Any, Callable, List, Optional, Sequence
+from typing import Any, Callable, Dict, List, Optional, Sequence
 
 from .events import EventEmitter
 
@@ -65,11 +70,12 @@ class StateScope:
     leaves the state that created it.
     """
 
-    __slots__ = ("_fsm", "_disposers", "_active")
+    __slots__ = ("_fsm", "_listeners", "_timers", "_active")
 
     def __init__(self, fsm: "FSM") -> None:
         self._fsm = fsm
-        self._disposers: List[Callable[[], None]] = []
+        self._listeners: Optional[List] = None   # flat [em, evt, cb, ...]
+        self._timers: Optional[List] = None      # cancellables
         self._active = True
 
     # -- queries ------------------------------------------------------
@@ -82,14 +88,25 @@ class StateScope:
         if not self._active:
             raise FSMError("S.on() used on exited state scope")
         emitter.on(event, cb)
-        self._disposers.append(lambda: emitter.remove_listener(event, cb))
+        ls = self._listeners
+        if ls is None:
+            ls = self._listeners = []
+        ls.append(emitter)
+        ls.append(event)
+        ls.append(cb)
+
+    def _add_timer(self, handle: Any) -> None:
+        ts = self._timers
+        if ts is None:
+            ts = self._timers = []
+        ts.append(handle)
 
     def timeout(self, ms: float, cb: Callable[[], None]) -> None:
         """Run cb after `ms` milliseconds unless the state is exited first."""
         if not self._active:
             raise FSMError("S.timeout() used on exited state scope")
-        handle = self._fsm._loop.call_later(ms / 1000.0, self._guarded(cb))
-        self._disposers.append(handle.cancel)
+        self._add_timer(self._fsm._loop.call_later(
+            ms / 1000.0, self._guarded(cb)))
 
     def interval(self, ms: float, cb: Callable[[], None]) -> None:
         if not self._active:
@@ -105,18 +122,21 @@ class StateScope:
 
         state["h"] = self._fsm._loop.call_later(ms / 1000.0, tick)
 
-        def dispose() -> None:
-            state["stop"] = True
-            if state["h"] is not None:
-                state["h"].cancel()
+        class _IntervalHandle:
+            __slots__ = ()
 
-        self._disposers.append(dispose)
+            @staticmethod
+            def cancel() -> None:
+                state["stop"] = True
+                if state["h"] is not None:
+                    state["h"].cancel()
+
+        self._add_timer(_IntervalHandle)
 
     def immediate(self, cb: Callable[[], None]) -> None:
         if not self._active:
             raise FSMError("S.immediate() used on exited state scope")
-        handle = self._fsm._loop.call_soon(self._guarded(cb))
-        self._disposers.append(handle.cancel)
+        self._add_timer(self._fsm._loop.call_soon(self._guarded(cb)))
 
     def callback(self, cb: Callable) -> Callable:
         """Wrap cb so it becomes a no-op once the state has been exited."""
@@ -136,7 +156,7 @@ class StateScope:
 
     # -- transitions ---------------------------------------------------
     def valid_transitions(self, states: Sequence[str]) -> None:
-        self._fsm._fsm_valid = list(states)
+        self._fsm._fsm_valid = states
 
     def goto_state(self, state: str) -> None:
         if not self._active:
@@ -155,9 +175,16 @@ class StateScope:
     # -- teardown ------------------------------------------------------
     def _dispose(self) -> None:
         self._active = False
-        disposers, self._disposers = self._disposers, []
-        for d in disposers:
-            d()
+        ls = self._listeners
+        if ls is not None:
+            self._listeners = None
+            for i in range(0, len(ls), 3):
+                ls[i].remove_listener(ls[i + 1], ls[i + 2])
+        ts = self._timers
+        if ts is not None:
+            self._timers = None
+            for h in ts:
+                h.cancel()
 
 
 class FSM(EventEmitter):
@@ -179,13 +206,20 @@ class FSM(EventEmitter):
     #: keeps history visible in core dumps)
     HISTORY_LEN = 8
 
+    #: per-class cache of state name -> entry method (unbound)
+    _fsm_entry_cache: Dict[str, Callable] = {}
+
+    def __init_subclass__(cls, **kw: Any) -> None:
+        super().__init_subclass__(**kw)
+        cls._fsm_entry_cache = {}
+
     def __init__(self, initial_state: str,
                  loop: Optional[asyncio.AbstractEventLoop] = None) -> None:
         super().__init__()
         self._loop = get_loop(loop)
         self._fsm_state: Optional[str] = None
         self._fsm_scope: Optional[StateScope] = None
-        self._fsm_valid: Optional[List[str]] = None
+        self._fsm_valid: Optional[Sequence[str]] = None
         self._fsm_entering = False
         self._fsm_pending: Optional[str] = None
         self._fsm_emit_queue: List[str] = []
@@ -210,10 +244,11 @@ class FSM(EventEmitter):
 
     # -- transitions ---------------------------------------------------
     def goto_state(self, state: str) -> None:
-        if self._fsm_valid is not None and state not in self._fsm_valid:
+        valid = self._fsm_valid
+        if valid is not None and state not in valid:
             raise FSMError(
                 "%s: invalid transition %r -> %r (valid: %r)"
-                % (type(self).__name__, self._fsm_state, state, self._fsm_valid)
+                % (type(self).__name__, self._fsm_state, state, valid)
             )
         if self._fsm_entering:
             # Requested while this FSM's entry function is still running:
@@ -228,8 +263,23 @@ class FSM(EventEmitter):
             return
         self._enter_loop(state)
 
+    def _entry_for(self, state: str) -> Callable:
+        cls = type(self)
+        cache = cls._fsm_entry_cache
+        entry = cache.get(state)
+        if entry is None:
+            entry = getattr(cls, "state_" + state.replace(".", "_"), None)
+            if entry is None:
+                raise FSMError(
+                    "%s has no state-entry function for %r"
+                    % (cls.__name__, state)
+                )
+            cache[state] = entry
+        return entry
+
     def _enter_loop(self, state: str) -> None:
         next_state: Optional[str] = state
+        hist = self._fsm_history
         while next_state is not None:
             target = next_state
             next_state = None
@@ -237,29 +287,25 @@ class FSM(EventEmitter):
                 self._fsm_scope._dispose()
             self._fsm_valid = None
             self._fsm_state = target
-            self._fsm_history.append(target)
-            if len(self._fsm_history) > self.HISTORY_LEN:
-                del self._fsm_history[0]
+            hist.append(target)
+            if len(hist) > self.HISTORY_LEN:
+                del hist[0]
             scope = StateScope(self)
             self._fsm_scope = scope
-            entry = getattr(self, "state_" + target.replace(".", "_"), None)
-            if entry is None:
-                raise FSMError(
-                    "%s has no state-entry function for %r"
-                    % (type(self).__name__, target)
-                )
+            entry = self._entry_for(target)
             self._fsm_entering = True
             try:
-                entry(scope)
+                entry(self, scope)
             finally:
                 self._fsm_entering = False
                 pend, self._fsm_pending = self._fsm_pending, None
             self._queue_state_changed(target)
             if pend is not None:
-                if self._fsm_valid is not None and pend not in self._fsm_valid:
+                valid = self._fsm_valid
+                if valid is not None and pend not in valid:
                     raise FSMError(
                         "%s: invalid transition %r -> %r (valid: %r)"
-                        % (type(self).__name__, target, pend, self._fsm_valid)
+                        % (type(self).__name__, target, pend, valid)
                     )
                 next_state = pend
 
@@ -272,6 +318,7 @@ class FSM(EventEmitter):
 
     def _flush_state_changed(self) -> None:
         self._fsm_emit_scheduled = False
-        while self._fsm_emit_queue:
-            st = self._fsm_emit_queue.pop(0)
+        q = self._fsm_emit_queue
+        while q:
+            st = q.pop(0)
             self.emit("stateChanged", st)

@@ -132,6 +132,8 @@ class ConnectionPool(FSM):
             service=options.get("service"),
             pool=self.p_uuid,
         )
+        # shared child logger for claim handles (hot path)
+        self.p_claim_log = self.p_log.child(component="ClaimHandle")
 
         self.p_collector = mod_utils.create_error_metrics(options)
 
@@ -676,7 +678,8 @@ class ConnectionPool(FSM):
             "pool": self,
             "claimStack": stack,
             "callback": cb,
-            "log": self.p_log,
+            "log": self.p_claim_log,
+            "_logReady": True,
             "claimTimeout": timeout,
             "loop": self._loop,
         })

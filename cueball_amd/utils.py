@@ -78,10 +78,11 @@ _DISABLED_STACK = [
 
 
 def maybe_capture_stack_trace() -> List[str]:
-    """Real stack frames if enabled, else a 2-frame placeholder
-    (lib/utils.js:106-115)."""
+    """Real stack frames if enabled, else a shared 2-frame placeholder
+    (lib/utils.js:106-115; off by default "for performance" — the
+    shared list keeps the disabled path allocation-free)."""
     if not _STACK_TRACES_ENABLED:
-        return list(_DISABLED_STACK)
+        return _DISABLED_STACK
     frames = traceback.extract_stack()[:-1]
     return ["%s (%s:%d)" % (f.name, f.filename, f.lineno) for f in frames]
 
