@@ -11,6 +11,7 @@ CoDel / TTL logic can be tested in milliseconds of real time.
 from __future__ import annotations
 
 import asyncio
+import json
 import selectors
 from typing import Any, Callable, Dict, List, Optional, Tuple
 
@@ -200,6 +201,108 @@ class DummyResolver(EventEmitter):
         self._last_error = err
         self.state = "failed"
         self.emit("stateChanged", "failed")
+
+
+class MockHttpServer:
+    """Tiny asyncio HTTP/1.1 server for agent tests and benchmarks
+    (stands in for the reference's local restify servers,
+    test/agent.test.js:31-44).
+
+    Routes: any path -> 200 JSON {"path":..., "count": N}; ``/ping`` ->
+    200 "pong"; ``/err500`` -> 500; ``/close`` -> 200 + Connection:
+    close.  ``broken=True`` accepts then immediately destroys
+    connections.  Counts requests per connection to verify keep-alive
+    reuse.
+    """
+
+    def __init__(self, broken: bool = False, tls_context=None) -> None:
+        self.broken = broken
+        self.tls_context = tls_context
+        self.port: Optional[int] = None
+        self.request_count = 0
+        self.ping_count = 0
+        self.conn_count = 0
+        self.requests_per_conn: List[int] = []
+        self._server: Optional[asyncio.AbstractServer] = None
+
+    async def start(self, port: int = 0) -> int:
+        self._server = await asyncio.start_server(
+            self._handle, "127.0.0.1", port, ssl=self.tls_context)
+        self.port = self._server.sockets[0].getsockname()[1]
+        return self.port
+
+    def stop(self) -> None:
+        if self._server is not None:
+            self._server.close()
+            self._server = None
+
+    async def _handle(self, reader: asyncio.StreamReader,
+                      writer: asyncio.StreamWriter) -> None:
+        self.conn_count += 1
+        my_count = 0
+        self.requests_per_conn.append(0)
+        slot = len(self.requests_per_conn) - 1
+        try:
+            if self.broken:
+                writer.close()
+                return
+            while True:
+                line = await reader.readline()
+                if not line or line == b"\r\n":
+                    if not line:
+                        break
+                    continue
+                parts = line.decode("latin-1").split()
+                if len(parts) < 3:
+                    break
+                method, path = parts[0], parts[1]
+                headers = {}
+                while True:
+                    hline = await reader.readline()
+                    if hline in (b"\r\n", b"", b"\n"):
+                        break
+                    name, _, value = hline.decode("latin-1").partition(":")
+                    headers[name.strip().lower()] = value.strip()
+                clen = int(headers.get("content-length", "0"))
+                if clen:
+                    await reader.readexactly(clen)
+
+                self.request_count += 1
+                my_count += 1
+                self.requests_per_conn[slot] = my_count
+
+                close = headers.get("connection", "").lower() == "close"
+                if path == "/ping":
+                    self.ping_count += 1
+                    body = b"pong"
+                    status = b"200 OK"
+                elif path == "/err500":
+                    body = b"boom"
+                    status = b"500 Internal Server Error"
+                elif path == "/close":
+                    body = b"bye"
+                    status = b"200 OK"
+                    close = True
+                else:
+                    body = json.dumps({"path": path,
+                                       "count": my_count}).encode()
+                    status = b"200 OK"
+                conn_hdr = b"close" if close else b"keep-alive"
+                writer.write(b"HTTP/1.1 " + status + b"\r\n"
+                             b"Content-Length: " +
+                             str(len(body)).encode() + b"\r\n"
+                             b"Connection: " + conn_hdr + b"\r\n\r\n" + body)
+                await writer.drain()
+                if close:
+                    break
+        except (ConnectionResetError, asyncio.IncompleteReadError,
+                BrokenPipeError):
+            pass
+        finally:
+            try:
+                writer.close()
+            except Exception:
+                pass
 
 
 class MockDnsServer:

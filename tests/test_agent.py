@@ -1,0 +1,272 @@
+"""HttpAgent / HttpsAgent tests (port of reference test/agent.test.js).
+
+Real localhost servers (MockHttpServer), fixed-IP pools, keep-alive
+reuse, initialDomains, create_pool args, the pinger, connection-refused
+and broken-server error paths, and HTTPS with a self-signed cert.
+"""
+
+import asyncio
+import json
+import os
+import ssl
+
+import pytest
+
+from cueball_amd.agent import HttpAgent, HttpsAgent
+from cueball_amd.testing import MockHttpServer
+
+RECOVERY = {"default": {"timeout": 2000, "retries": 2, "delay": 50,
+                        "maxDelay": 200}}
+
+FIXTURES = os.path.join(os.path.dirname(__file__), "fixtures")
+
+
+def run(coro):
+    loop = asyncio.new_event_loop()
+    try:
+        return loop.run_until_complete(coro)
+    finally:
+        loop.close()
+
+
+async def stop_agent(agent):
+    fut = asyncio.get_running_loop().create_future()
+    agent.stop(lambda err: fut.set_result(err))
+    await fut
+
+
+def test_basic_agent_usage_fixed_ip():
+    async def body():
+        srv = MockHttpServer()
+        await srv.start()
+        agent = HttpAgent({
+            "defaultPort": srv.port,
+            "recovery": RECOVERY,
+            "spares": 1,
+            "maximum": 1,  # exactly one conn => reuse is observable
+        })
+        resp = await agent.request_async("127.0.0.1", "GET", "/hello")
+        assert resp.status_code == 200
+        data = json.loads(resp.body)
+        assert data["path"] == "/hello"
+
+        # second request reuses the kept-alive connection
+        resp2 = await agent.request_async("127.0.0.1", "GET", "/again")
+        assert resp2.status_code == 200
+        assert json.loads(resp2.body)["count"] == 2
+        assert srv.conn_count == 1
+
+        await stop_agent(agent)
+        srv.stop()
+
+    run(body())
+
+
+def test_agent_initial_domains():
+    async def body():
+        srv = MockHttpServer()
+        await srv.start()
+        agent = HttpAgent({
+            "defaultPort": srv.port,
+            "recovery": RECOVERY,
+            "spares": 1,
+            "maximum": 2,
+            "initialDomains": ["127.0.0.1"],
+        })
+        # the pool exists before any request
+        assert agent.get_pool("127.0.0.1") is not None
+        await asyncio.sleep(0.2)
+        # spare connection pre-opened
+        assert srv.conn_count >= 1
+        resp = await agent.request_async("127.0.0.1", "GET", "/x")
+        assert resp.status_code == 200
+        await stop_agent(agent)
+        srv.stop()
+
+    run(body())
+
+
+def test_agent_create_pool_args():
+    async def body():
+        srv = MockHttpServer()
+        await srv.start()
+        agent = HttpAgent({
+            "defaultPort": 1,  # wrong on purpose; create_pool overrides
+            "recovery": RECOVERY,
+            "spares": 1,
+            "maximum": 2,
+        })
+        agent.create_pool("127.0.0.1", {"port": srv.port})
+        assert agent.get_pool("127.0.0.1") is not None
+        with pytest.raises(Exception):
+            agent.create_pool("127.0.0.1")
+        resp = await agent.request_async("127.0.0.1", "GET", "/y")
+        assert resp.status_code == 200
+        await stop_agent(agent)
+        srv.stop()
+
+    run(body())
+
+
+def test_agent_pinger():
+    async def body():
+        srv = MockHttpServer()
+        await srv.start()
+        agent = HttpAgent({
+            "defaultPort": srv.port,
+            "recovery": RECOVERY,
+            "spares": 1,
+            "maximum": 2,
+            "ping": "/ping",
+            "pingInterval": 150,
+        })
+        resp = await agent.request_async("127.0.0.1", "GET", "/warm")
+        assert resp.status_code == 200
+        await asyncio.sleep(0.8)
+        assert srv.ping_count >= 2  # idle conn pinged repeatedly
+        await stop_agent(agent)
+        srv.stop()
+
+    run(body())
+
+
+def test_agent_connection_refused():
+    async def body():
+        # a port with nothing listening
+        probe = MockHttpServer()
+        await probe.start()
+        dead_port = probe.port
+        probe.stop()
+        await asyncio.sleep(0.05)
+
+        agent = HttpAgent({
+            "defaultPort": dead_port,
+            "recovery": {"default": {"timeout": 500, "retries": 1,
+                                     "delay": 10}},
+            "spares": 1,
+            "maximum": 1,
+            "errorOnEmpty": False,
+        })
+        with pytest.raises(Exception):
+            await asyncio.wait_for(
+                agent.request_async("127.0.0.1", "GET", "/"), timeout=10)
+        await stop_agent(agent)
+
+    run(body())
+
+
+def test_agent_on_broken_server():
+    async def body():
+        srv = MockHttpServer(broken=True)
+        await srv.start()
+        agent = HttpAgent({
+            "defaultPort": srv.port,
+            "recovery": {"default": {"timeout": 500, "retries": 1,
+                                     "delay": 10}},
+            "spares": 1,
+            "maximum": 1,
+        })
+        with pytest.raises(Exception):
+            await asyncio.wait_for(
+                agent.request_async("127.0.0.1", "GET", "/"), timeout=10)
+        await stop_agent(agent)
+        srv.stop()
+
+    run(body())
+
+
+def test_https_agent():
+    async def body():
+        server_ctx = ssl.SSLContext(ssl.PROTOCOL_TLS_SERVER)
+        server_ctx.load_cert_chain(
+            os.path.join(FIXTURES, "test_cert.pem"),
+            os.path.join(FIXTURES, "test_key.pem"))
+        srv = MockHttpServer(tls_context=server_ctx)
+        await srv.start()
+
+        client_ctx = ssl.SSLContext(ssl.PROTOCOL_TLS_CLIENT)
+        client_ctx.load_verify_locations(
+            os.path.join(FIXTURES, "test_cert.pem"))
+        client_ctx.check_hostname = False
+
+        agent = HttpsAgent({
+            "defaultPort": srv.port,
+            "recovery": RECOVERY,
+            "spares": 1,
+            "maximum": 2,
+        })
+        agent.create_pool("127.0.0.1", {"port": srv.port,
+                                        "ssl_context": client_ctx})
+        resp = await agent.request_async("127.0.0.1", "GET", "/tls")
+        assert resp.status_code == 200
+        assert json.loads(resp.body)["path"] == "/tls"
+        await stop_agent(agent)
+        srv.stop()
+
+    run(body())
+
+
+def test_agent_stopped_rejects_requests():
+    async def body():
+        srv = MockHttpServer()
+        await srv.start()
+        agent = HttpAgent({
+            "defaultPort": srv.port,
+            "recovery": RECOVERY,
+            "spares": 1,
+            "maximum": 2,
+        })
+        await agent.request_async("127.0.0.1", "GET", "/")
+        await stop_agent(agent)
+        with pytest.raises(Exception):
+            await agent.request_async("127.0.0.1", "GET", "/")
+        srv.stop()
+
+    run(body())
+
+
+def test_connection_close_header_not_reused():
+    async def body():
+        srv = MockHttpServer()
+        await srv.start()
+        agent = HttpAgent({
+            "defaultPort": srv.port,
+            "recovery": RECOVERY,
+            "spares": 1,
+            "maximum": 2,
+        })
+        resp = await agent.request_async("127.0.0.1", "GET", "/close")
+        assert resp.status_code == 200
+        assert resp.body == b"bye"
+        await asyncio.sleep(0.3)
+        resp2 = await agent.request_async("127.0.0.1", "GET", "/n")
+        assert resp2.status_code == 200
+        # server closed after /close: second request used a new conn
+        assert srv.conn_count >= 2
+        await stop_agent(agent)
+        srv.stop()
+
+    run(body())
+
+
+def test_pinger_5xx_closes_connection():
+    async def body():
+        srv = MockHttpServer()
+        await srv.start()
+        agent = HttpAgent({
+            "defaultPort": srv.port,
+            "recovery": RECOVERY,
+            "spares": 1,
+            "maximum": 2,
+            "ping": "/err500",
+            "pingInterval": 120,
+        })
+        await agent.request_async("127.0.0.1", "GET", "/warm")
+        n0 = srv.conn_count
+        await asyncio.sleep(0.8)
+        # every ping 500s -> connection closed and re-made repeatedly
+        assert srv.conn_count > n0
+        await stop_agent(agent)
+        srv.stop()
+
+    run(body())
