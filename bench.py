@@ -391,12 +391,80 @@ async def scenario_codel(args, results):
     await asyncio.sleep(0.1)
 
 
+async def scenario_agent(args, results):
+    """HttpAgent keep-alive over 8 local HTTP backends, 1000 concurrent
+    GETs per step (BASELINE config #3)."""
+    from cueball_amd.agent import HttpAgent
+    from cueball_amd.testing import DummyResolver, MockHttpServer
+
+    loop = asyncio.get_running_loop()
+    servers = []
+    for _ in range(8):
+        s = MockHttpServer()
+        await s.start()
+        servers.append(s)
+
+    resolver = DummyResolver()
+    agent = HttpAgent({
+        "defaultPort": servers[0].port,
+        "recovery": {"default": {"timeout": 2000, "retries": 3,
+                                 "delay": 100, "maxDelay": 2000}},
+        "spares": 8,
+        "maximum": 32,
+        "loop": loop,
+    })
+    agent.create_pool("svc.bench", {"resolver": resolver})
+    resolver.start()
+    for i, s in enumerate(servers):
+        resolver.add("b%d" % i, {"address": "127.0.0.1", "port": s.port,
+                                 "name": "b%d" % i})
+    pool = agent.get_pool("svc.bench")
+    t_deadline = time.monotonic() + 15
+    while pool.get_stats()["idleConnections"] < 8 and \
+            time.monotonic() < t_deadline:
+        await asyncio.sleep(0.01)
+
+    concurrent = 1000
+    lat = []
+
+    async def one_get(i):
+        t0 = loop.time()
+        resp = await agent.request_async("svc.bench", "GET", "/%d" % i)
+        lat.append(loop.time() - t0)
+        assert resp.status_code == 200
+
+    async def step():
+        await asyncio.gather(*[one_get(i) for i in range(concurrent)])
+
+    for _ in range(args.warmup):
+        await step()
+    lat.clear()
+    barrier()
+    t0 = time.perf_counter()
+    for _ in range(args.steps):
+        await step()
+    t1 = time.perf_counter()
+    barrier()
+    results["elapsed"] = t1 - t0
+    results["ops"] = args.steps * concurrent
+    results["lat_p50"] = statistics.median(lat) * 1000 if lat else None
+    results["lat_p99"] = (statistics.quantiles(lat, n=100)[98] * 1000
+                          if len(lat) >= 100 else None)
+    fut = loop.create_future()
+    agent.stop(lambda e: fut.set_result(None))
+    await fut
+    for s in servers:
+        s.stop()
+    await asyncio.sleep(0.1)
+
+
 SCENARIOS = {
     "headline": scenario_headline,
     "static1": scenario_static1,
     "dns": scenario_dns,
     "cset": scenario_cset,
     "codel": scenario_codel,
+    "agent": scenario_agent,
 }
 
 
