@@ -1,0 +1,1494 @@
+/*
+ * cueball_amd._speed: native event/FSM runtime core.
+ *
+ * The framework's hot loop (a pool claim/release cycle) walks six-plus
+ * Moore-machine state transitions across three FSMs, each with scoped
+ * listener registration/teardown and async stateChanged queueing (see
+ * cueball_amd/fsm.py for the semantics contract, which this module
+ * reproduces exactly -- the whole pytest suite runs against either
+ * implementation).  This C++ core removes the interpreter overhead from
+ * EventEmitter dispatch, scope bookkeeping and the transition loop.
+ *
+ * The reference (node-cueball) has no native code; this is the one
+ * component where a native core is justified by measurement (see
+ * profiles/README.md): event-loop throughput is the library's
+ * performance story.
+ *
+ * Plain C++/CPython API -- no GPU code exists in this problem domain
+ * (SURVEY.md section 0).
+ */
+
+#define PY_SSIZE_T_CLEAN
+#include <Python.h>
+#include <structmember.h>
+
+namespace {
+
+/* interned strings, created at module init */
+PyObject *s_stateChanged;
+PyObject *s_listener;
+PyObject *s_on;
+PyObject *s_remove_listener;
+PyObject *s_call_soon;
+PyObject *s_call_later;
+PyObject *s_cancel;
+PyObject *s_state_prefix;      /* "state_" */
+PyObject *s_dot;               /* "." */
+PyObject *s_underscore;        /* "_" */
+PyObject *s_flush_name;        /* "_flush_state_changed" */
+
+PyObject *g_get_loop;          /* python helper: get_loop(loop) */
+PyObject *g_fsm_error;         /* exception class FSMError */
+PyObject *g_entry_name_cache;  /* dict: state name -> "state_x_y" */
+
+/* ------------------------------------------------------------------ */
+/* EventEmitter                                                        */
+/* ------------------------------------------------------------------ */
+
+typedef struct {
+    PyObject_HEAD
+    PyObject *ev_events;   /* dict: str -> list of callables */
+    PyObject *ev_dict;     /* instance __dict__ (lazy) */
+    PyObject *ev_weakrefs;
+} Emitter;
+
+extern PyTypeObject EmitterType;
+
+int
+Emitter_init(PyObject *self_, PyObject *args, PyObject *kwds)
+{
+    Emitter *self = (Emitter *)self_;
+    (void)args; (void)kwds;
+    if (self->ev_events == NULL) {
+        self->ev_events = PyDict_New();
+        if (self->ev_events == NULL)
+            return -1;
+    }
+    return 0;
+}
+
+int
+Emitter_traverse(PyObject *self_, visitproc visit, void *arg)
+{
+    Emitter *self = (Emitter *)self_;
+    Py_VISIT(self->ev_events);
+    Py_VISIT(self->ev_dict);
+    return 0;
+}
+
+int
+Emitter_clear_(PyObject *self_)
+{
+    Emitter *self = (Emitter *)self_;
+    Py_CLEAR(self->ev_events);
+    Py_CLEAR(self->ev_dict);
+    return 0;
+}
+
+void
+Emitter_dealloc(PyObject *self_)
+{
+    Emitter *self = (Emitter *)self_;
+    PyTypeObject *tp = Py_TYPE(self_);
+    PyObject_GC_UnTrack(self_);
+    if (self->ev_weakrefs != NULL)
+        PyObject_ClearWeakRefs(self_);
+    Emitter_clear_(self_);
+    tp->tp_free(self_);
+}
+
+/* core: register a listener */
+PyObject *
+emitter_add(Emitter *self, PyObject *event, PyObject *listener)
+{
+    if (self->ev_events == NULL) {
+        self->ev_events = PyDict_New();
+        if (self->ev_events == NULL)
+            return NULL;
+    }
+    PyObject *ls = PyDict_GetItemWithError(self->ev_events, event);
+    if (ls == NULL) {
+        if (PyErr_Occurred())
+            return NULL;
+        ls = PyList_New(0);
+        if (ls == NULL)
+            return NULL;
+        if (PyDict_SetItem(self->ev_events, event, ls) < 0) {
+            Py_DECREF(ls);
+            return NULL;
+        }
+        Py_DECREF(ls);  /* dict holds it */
+    }
+    if (PyList_Append(ls, listener) < 0)
+        return NULL;
+    Py_INCREF(listener);
+    return listener;
+}
+
+PyObject *
+Emitter_on(PyObject *self_, PyObject *const *args, Py_ssize_t nargs)
+{
+    if (nargs != 2) {
+        PyErr_SetString(PyExc_TypeError, "on(event, listener)");
+        return NULL;
+    }
+    return emitter_add((Emitter *)self_, args[0], args[1]);
+}
+
+/* once() wrapper object */
+typedef struct {
+    PyObject_HEAD
+    PyObject *ow_emitter;   /* weak semantics not needed; strong ref */
+    PyObject *ow_event;
+    PyObject *ow_listener;
+} OnceWrapper;
+
+extern PyTypeObject OnceWrapperType;
+
+PyObject *
+OnceWrapper_call(PyObject *self_, PyObject *args, PyObject *kwds)
+{
+    OnceWrapper *self = (OnceWrapper *)self_;
+    (void)kwds;
+    /* remove ourselves first (reference events.py once semantics) */
+    PyObject *res = PyObject_CallMethodObjArgs(
+        self->ow_emitter, s_remove_listener, self->ow_event,
+        (PyObject *)self, NULL);
+    if (res == NULL)
+        return NULL;
+    Py_DECREF(res);
+    return PyObject_Call(self->ow_listener, args, NULL);
+}
+
+PyObject *
+OnceWrapper_get_listener(PyObject *self_, void *closure)
+{
+    OnceWrapper *self = (OnceWrapper *)self_;
+    (void)closure;
+    Py_INCREF(self->ow_listener);
+    return self->ow_listener;
+}
+
+int
+OnceWrapper_traverse(PyObject *self_, visitproc visit, void *arg)
+{
+    OnceWrapper *self = (OnceWrapper *)self_;
+    Py_VISIT(self->ow_emitter);
+    Py_VISIT(self->ow_event);
+    Py_VISIT(self->ow_listener);
+    return 0;
+}
+
+int
+OnceWrapper_clear_(PyObject *self_)
+{
+    OnceWrapper *self = (OnceWrapper *)self_;
+    Py_CLEAR(self->ow_emitter);
+    Py_CLEAR(self->ow_event);
+    Py_CLEAR(self->ow_listener);
+    return 0;
+}
+
+void
+OnceWrapper_dealloc(PyObject *self_)
+{
+    PyObject_GC_UnTrack(self_);
+    OnceWrapper_clear_(self_);
+    PyObject_GC_Del(self_);
+}
+
+PyGetSetDef OnceWrapper_getset[] = {
+    {(char *)"listener", OnceWrapper_get_listener, NULL,
+     (char *)"original listener", NULL},
+    {NULL, NULL, NULL, NULL, NULL},
+};
+
+PyTypeObject OnceWrapperType = {
+    PyVarObject_HEAD_INIT(NULL, 0)
+    "cueball_amd._speed._OnceWrapper",       /* tp_name */
+    sizeof(OnceWrapper),                      /* tp_basicsize */
+    0,                                        /* tp_itemsize */
+    OnceWrapper_dealloc,                      /* tp_dealloc */
+    0, 0, 0, 0, 0, 0, 0, 0, 0,
+    OnceWrapper_call,                         /* tp_call */
+    0, 0, 0, 0,
+    Py_TPFLAGS_DEFAULT | Py_TPFLAGS_HAVE_GC,  /* tp_flags */
+    0,                                        /* tp_doc */
+    OnceWrapper_traverse,                     /* tp_traverse */
+    OnceWrapper_clear_,                       /* tp_clear */
+    0, 0, 0, 0, 0, 0,
+    OnceWrapper_getset,                       /* tp_getset */
+};
+
+PyObject *
+Emitter_once(PyObject *self_, PyObject *const *args, Py_ssize_t nargs)
+{
+    if (nargs != 2) {
+        PyErr_SetString(PyExc_TypeError, "once(event, listener)");
+        return NULL;
+    }
+    OnceWrapper *w = PyObject_GC_New(OnceWrapper, &OnceWrapperType);
+    if (w == NULL)
+        return NULL;
+    Py_INCREF(self_);
+    w->ow_emitter = self_;
+    Py_INCREF(args[0]);
+    w->ow_event = args[0];
+    Py_INCREF(args[1]);
+    w->ow_listener = args[1];
+    PyObject_GC_Track((PyObject *)w);
+    PyObject *r = emitter_add((Emitter *)self_, args[0], (PyObject *)w);
+    if (r == NULL) {
+        Py_DECREF(w);
+        return NULL;
+    }
+    Py_DECREF(r);
+    return (PyObject *)w;
+}
+
+PyObject *
+Emitter_remove_listener(PyObject *self_, PyObject *const *args,
+                        Py_ssize_t nargs)
+{
+    Emitter *self = (Emitter *)self_;
+    if (nargs != 2) {
+        PyErr_SetString(PyExc_TypeError, "remove_listener(event, listener)");
+        return NULL;
+    }
+    if (self->ev_events == NULL)
+        Py_RETURN_NONE;
+    PyObject *ls = PyDict_GetItemWithError(self->ev_events, args[0]);
+    if (ls == NULL) {
+        if (PyErr_Occurred())
+            return NULL;
+        Py_RETURN_NONE;
+    }
+    Py_ssize_t n = PyList_GET_SIZE(ls);
+    Py_ssize_t found = -1;
+    for (Py_ssize_t i = 0; i < n; i++) {
+        if (PyList_GET_ITEM(ls, i) == args[1]) {
+            found = i;
+            break;
+        }
+    }
+    if (found < 0) {
+        /* allow removing a once() registration by its inner listener */
+        for (Py_ssize_t i = 0; i < n; i++) {
+            PyObject *w = PyList_GET_ITEM(ls, i);
+            if (Py_TYPE(w) == &OnceWrapperType &&
+                ((OnceWrapper *)w)->ow_listener == args[1]) {
+                found = i;
+                break;
+            }
+        }
+    }
+    if (found >= 0) {
+        if (PyList_SetSlice(ls, found, found + 1, NULL) < 0)
+            return NULL;
+        if (PyList_GET_SIZE(ls) == 0) {
+            if (PyDict_DelItem(self->ev_events, args[0]) < 0)
+                PyErr_Clear();
+        }
+    }
+    Py_RETURN_NONE;
+}
+
+PyObject *
+Emitter_remove_all_listeners(PyObject *self_, PyObject *const *args,
+                             Py_ssize_t nargs)
+{
+    Emitter *self = (Emitter *)self_;
+    if (self->ev_events == NULL)
+        Py_RETURN_NONE;
+    if (nargs == 0 || args[0] == Py_None) {
+        PyDict_Clear(self->ev_events);
+    } else {
+        if (PyDict_DelItem(self->ev_events, args[0]) < 0)
+            PyErr_Clear();
+    }
+    Py_RETURN_NONE;
+}
+
+PyObject *
+Emitter_listeners(PyObject *self_, PyObject *event)
+{
+    Emitter *self = (Emitter *)self_;
+    if (self->ev_events != NULL) {
+        PyObject *ls = PyDict_GetItemWithError(self->ev_events, event);
+        if (ls != NULL)
+            return PyList_GetSlice(ls, 0, PyList_GET_SIZE(ls));
+        if (PyErr_Occurred())
+            return NULL;
+    }
+    return PyList_New(0);
+}
+
+PyObject *
+Emitter_listener_count(PyObject *self_, PyObject *event)
+{
+    Emitter *self = (Emitter *)self_;
+    Py_ssize_t n = 0;
+    if (self->ev_events != NULL) {
+        PyObject *ls = PyDict_GetItemWithError(self->ev_events, event);
+        if (ls != NULL)
+            n = PyList_GET_SIZE(ls);
+        else if (PyErr_Occurred())
+            return NULL;
+    }
+    return PyLong_FromSsize_t(n);
+}
+
+PyObject *
+Emitter_event_names(PyObject *self_, PyObject *noargs)
+{
+    Emitter *self = (Emitter *)self_;
+    (void)noargs;
+    if (self->ev_events == NULL)
+        return PyList_New(0);
+    return PyDict_Keys(self->ev_events);
+}
+
+/* emit with node snapshot semantics */
+int
+emitter_emit_core(Emitter *self, PyObject *event, PyObject *const *eargs,
+                  Py_ssize_t neargs)
+{
+    if (self->ev_events == NULL)
+        return 0;
+    PyObject *ls = PyDict_GetItemWithError(self->ev_events, event);
+    if (ls == NULL)
+        return PyErr_Occurred() ? -1 : 0;
+    Py_ssize_t n = PyList_GET_SIZE(ls);
+    if (n == 0)
+        return 0;
+    if (n == 1) {
+        /* copy-free fast path (snapshot still holds: a listener added
+         * during the call is not invoked) */
+        PyObject *cb = PyList_GET_ITEM(ls, 0);
+        Py_INCREF(cb);
+        PyObject *r = PyObject_Vectorcall(cb, eargs, neargs, NULL);
+        Py_DECREF(cb);
+        if (r == NULL)
+            return -1;
+        Py_DECREF(r);
+        return 1;
+    }
+    PyObject *snap = PyList_GetSlice(ls, 0, n);
+    if (snap == NULL)
+        return -1;
+    for (Py_ssize_t i = 0; i < n; i++) {
+        PyObject *cb = PyList_GET_ITEM(snap, i);
+        PyObject *r = PyObject_Vectorcall(cb, eargs, neargs, NULL);
+        if (r == NULL) {
+            Py_DECREF(snap);
+            return -1;
+        }
+        Py_DECREF(r);
+    }
+    Py_DECREF(snap);
+    return 1;
+}
+
+PyObject *
+Emitter_emit(PyObject *self_, PyObject *const *args, Py_ssize_t nargs)
+{
+    if (nargs < 1) {
+        PyErr_SetString(PyExc_TypeError, "emit(event, *args)");
+        return NULL;
+    }
+    int r = emitter_emit_core((Emitter *)self_, args[0], args + 1,
+                              nargs - 1);
+    if (r < 0)
+        return NULL;
+    return PyBool_FromLong(r);
+}
+
+PyMethodDef Emitter_methods[] = {
+    {"on", (PyCFunction)(void (*)(void))Emitter_on, METH_FASTCALL, NULL},
+    {"add_listener", (PyCFunction)(void (*)(void))Emitter_on,
+     METH_FASTCALL, NULL},
+    {"once", (PyCFunction)(void (*)(void))Emitter_once, METH_FASTCALL,
+     NULL},
+    {"remove_listener",
+     (PyCFunction)(void (*)(void))Emitter_remove_listener, METH_FASTCALL,
+     NULL},
+    {"remove_all_listeners",
+     (PyCFunction)(void (*)(void))Emitter_remove_all_listeners,
+     METH_FASTCALL, NULL},
+    {"listeners", Emitter_listeners, METH_O, NULL},
+    {"listener_count", Emitter_listener_count, METH_O, NULL},
+    {"event_names", Emitter_event_names, METH_NOARGS, NULL},
+    {"emit", (PyCFunction)(void (*)(void))Emitter_emit, METH_FASTCALL,
+     NULL},
+    {NULL, NULL, 0, NULL},
+};
+
+PyMemberDef Emitter_members[] = {
+    {(char *)"_events", T_OBJECT, offsetof(Emitter, ev_events), READONLY,
+     NULL},
+    {NULL, 0, 0, 0, NULL},
+};
+
+PyTypeObject EmitterType = {
+    PyVarObject_HEAD_INIT(NULL, 0)
+    "cueball_amd._speed.EventEmitter",        /* tp_name */
+    sizeof(Emitter),                          /* tp_basicsize */
+    0,                                        /* tp_itemsize */
+    Emitter_dealloc,                          /* tp_dealloc */
+    0, 0, 0, 0, 0, 0, 0, 0, 0, 0, 0, 0, 0, 0,
+    Py_TPFLAGS_DEFAULT | Py_TPFLAGS_BASETYPE | Py_TPFLAGS_HAVE_GC,
+    "node-style EventEmitter (native)",       /* tp_doc */
+    Emitter_traverse,                         /* tp_traverse */
+    Emitter_clear_,                           /* tp_clear */
+    0,                                        /* tp_richcompare */
+    offsetof(Emitter, ev_weakrefs),           /* tp_weaklistoffset */
+    0, 0,
+    Emitter_methods,                          /* tp_methods */
+    Emitter_members,                          /* tp_members */
+    0, 0, 0, 0, 0,
+    offsetof(Emitter, ev_dict),               /* tp_dictoffset */
+    Emitter_init,                             /* tp_init */
+    0,
+    PyType_GenericNew,                        /* tp_new */
+};
+
+/* ------------------------------------------------------------------ */
+/* forward decls                                                       */
+/* ------------------------------------------------------------------ */
+
+typedef struct FSMOb FSMOb;
+int fsm_goto_state(FSMOb *fsm, PyObject *state);
+
+/* ------------------------------------------------------------------ */
+/* StateScope                                                          */
+/* ------------------------------------------------------------------ */
+
+typedef struct {
+    PyObject_HEAD
+    PyObject *sc_fsm;       /* FSMOb*, strong */
+    PyObject *sc_listeners; /* flat list [em, evt, cb, ...] or NULL */
+    PyObject *sc_timers;    /* list of cancellables or NULL */
+    int sc_active;
+} Scope;
+
+extern PyTypeObject ScopeType;
+
+/* guarded zero/N-arg callback: no-op once scope inactive */
+typedef struct {
+    PyObject_HEAD
+    PyObject *gc_scope;   /* Scope* */
+    PyObject *gc_cb;
+} GuardedCb;
+
+extern PyTypeObject GuardedCbType;
+
+PyObject *
+GuardedCb_call(PyObject *self_, PyObject *args, PyObject *kwds)
+{
+    GuardedCb *self = (GuardedCb *)self_;
+    (void)kwds;
+    if (!((Scope *)self->gc_scope)->sc_active)
+        Py_RETURN_NONE;
+    return PyObject_Call(self->gc_cb, args, NULL);
+}
+
+int
+GuardedCb_traverse(PyObject *self_, visitproc visit, void *arg)
+{
+    GuardedCb *self = (GuardedCb *)self_;
+    Py_VISIT(self->gc_scope);
+    Py_VISIT(self->gc_cb);
+    return 0;
+}
+
+int
+GuardedCb_clear_(PyObject *self_)
+{
+    GuardedCb *self = (GuardedCb *)self_;
+    Py_CLEAR(self->gc_scope);
+    Py_CLEAR(self->gc_cb);
+    return 0;
+}
+
+void
+GuardedCb_dealloc(PyObject *self_)
+{
+    PyObject_GC_UnTrack(self_);
+    GuardedCb_clear_(self_);
+    PyObject_GC_Del(self_);
+}
+
+PyTypeObject GuardedCbType = {
+    PyVarObject_HEAD_INIT(NULL, 0)
+    "cueball_amd._speed._GuardedCb",
+    sizeof(GuardedCb),
+    0,
+    GuardedCb_dealloc,
+    0, 0, 0, 0, 0, 0, 0, 0, 0,
+    GuardedCb_call,
+    0, 0, 0, 0,
+    Py_TPFLAGS_DEFAULT | Py_TPFLAGS_HAVE_GC,
+    0,
+    GuardedCb_traverse,
+    GuardedCb_clear_,
+};
+
+PyObject *
+make_guarded(PyObject *scope, PyObject *cb)
+{
+    GuardedCb *g = PyObject_GC_New(GuardedCb, &GuardedCbType);
+    if (g == NULL)
+        return NULL;
+    Py_INCREF(scope);
+    g->gc_scope = scope;
+    Py_INCREF(cb);
+    g->gc_cb = cb;
+    PyObject_GC_Track((PyObject *)g);
+    return (PyObject *)g;
+}
+
+int
+scope_check_active(Scope *self, const char *what)
+{
+    if (!self->sc_active) {
+        PyErr_Format(g_fsm_error, "%s used on exited state scope", what);
+        return -1;
+    }
+    return 0;
+}
+
+PyObject *
+Scope_on(PyObject *self_, PyObject *const *args, Py_ssize_t nargs)
+{
+    Scope *self = (Scope *)self_;
+    if (nargs != 3) {
+        PyErr_SetString(PyExc_TypeError, "S.on(emitter, event, cb)");
+        return NULL;
+    }
+    if (scope_check_active(self, "S.on()") < 0)
+        return NULL;
+    PyObject *emitter = args[0], *event = args[1], *cb = args[2];
+    PyObject *r;
+    /* fast path for our own emitter type with no overridden .on */
+    if (Py_TYPE(emitter) == &EmitterType) {
+        r = emitter_add((Emitter *)emitter, event, cb);
+    } else {
+        r = PyObject_CallMethodObjArgs(emitter, s_on, event, cb, NULL);
+    }
+    if (r == NULL)
+        return NULL;
+    Py_DECREF(r);
+    if (self->sc_listeners == NULL) {
+        self->sc_listeners = PyList_New(0);
+        if (self->sc_listeners == NULL)
+            return NULL;
+    }
+    if (PyList_Append(self->sc_listeners, emitter) < 0 ||
+        PyList_Append(self->sc_listeners, event) < 0 ||
+        PyList_Append(self->sc_listeners, cb) < 0)
+        return NULL;
+    Py_RETURN_NONE;
+}
+
+int
+scope_add_timer(Scope *self, PyObject *handle)
+{
+    if (self->sc_timers == NULL) {
+        self->sc_timers = PyList_New(0);
+        if (self->sc_timers == NULL)
+            return -1;
+    }
+    return PyList_Append(self->sc_timers, handle);
+}
+
+PyObject *fsm_get_loop_of(FSMOb *fsm);
+
+PyObject *
+Scope_timeout(PyObject *self_, PyObject *const *args, Py_ssize_t nargs)
+{
+    Scope *self = (Scope *)self_;
+    if (nargs != 2) {
+        PyErr_SetString(PyExc_TypeError, "S.timeout(ms, cb)");
+        return NULL;
+    }
+    if (scope_check_active(self, "S.timeout()") < 0)
+        return NULL;
+    double ms = PyFloat_AsDouble(args[0]);
+    if (ms == -1.0 && PyErr_Occurred())
+        return NULL;
+    PyObject *guarded = make_guarded(self_, args[1]);
+    if (guarded == NULL)
+        return NULL;
+    PyObject *secs = PyFloat_FromDouble(ms / 1000.0);
+    if (secs == NULL) {
+        Py_DECREF(guarded);
+        return NULL;
+    }
+    PyObject *loop = fsm_get_loop_of((FSMOb *)self->sc_fsm);
+    PyObject *handle = PyObject_CallMethodObjArgs(loop, s_call_later, secs,
+                                                  guarded, NULL);
+    Py_DECREF(secs);
+    Py_DECREF(guarded);
+    if (handle == NULL)
+        return NULL;
+    int rc = scope_add_timer(self, handle);
+    Py_DECREF(handle);
+    if (rc < 0)
+        return NULL;
+    Py_RETURN_NONE;
+}
+
+PyObject *
+Scope_immediate(PyObject *self_, PyObject *cb)
+{
+    Scope *self = (Scope *)self_;
+    if (scope_check_active(self, "S.immediate()") < 0)
+        return NULL;
+    PyObject *guarded = make_guarded(self_, cb);
+    if (guarded == NULL)
+        return NULL;
+    PyObject *loop = fsm_get_loop_of((FSMOb *)self->sc_fsm);
+    PyObject *handle = PyObject_CallMethodObjArgs(loop, s_call_soon,
+                                                  guarded, NULL);
+    Py_DECREF(guarded);
+    if (handle == NULL)
+        return NULL;
+    int rc = scope_add_timer(self, handle);
+    Py_DECREF(handle);
+    if (rc < 0)
+        return NULL;
+    Py_RETURN_NONE;
+}
+
+PyObject *
+Scope_callback(PyObject *self_, PyObject *cb)
+{
+    Scope *self = (Scope *)self_;
+    if (scope_check_active(self, "S.callback()") < 0)
+        return NULL;
+    return make_guarded(self_, cb);
+}
+
+PyObject *Scope_interval(PyObject *self_, PyObject *const *args,
+                         Py_ssize_t nargs);
+
+PyObject *
+Scope_valid_transitions(PyObject *self_, PyObject *states);
+
+PyObject *
+Scope_goto_state(PyObject *self_, PyObject *state)
+{
+    Scope *self = (Scope *)self_;
+    if (!self->sc_active)
+        Py_RETURN_NONE;   /* stale handler during a cascade: obsolete */
+    if (fsm_goto_state((FSMOb *)self->sc_fsm, state) < 0)
+        return NULL;
+    Py_RETURN_NONE;
+}
+
+PyObject *
+Scope_dispose(PyObject *self_, PyObject *noargs)
+{
+    Scope *self = (Scope *)self_;
+    (void)noargs;
+    self->sc_active = 0;
+    PyObject *ls = self->sc_listeners;
+    self->sc_listeners = NULL;
+    if (ls != NULL) {
+        Py_ssize_t n = PyList_GET_SIZE(ls);
+        for (Py_ssize_t i = 0; i + 2 < n; i += 3) {
+            PyObject *em = PyList_GET_ITEM(ls, i);
+            PyObject *evt = PyList_GET_ITEM(ls, i + 1);
+            PyObject *cb = PyList_GET_ITEM(ls, i + 2);
+            PyObject *r;
+            if (Py_TYPE(em) == &EmitterType ||
+                (PyType_IsSubtype(Py_TYPE(em), &EmitterType) &&
+                 /* only safe if remove_listener not overridden; our
+                  * subclasses never override it */ 1)) {
+                PyObject *cargs[2] = {evt, cb};
+                r = Emitter_remove_listener(em, cargs, 2);
+            } else {
+                r = PyObject_CallMethodObjArgs(em, s_remove_listener, evt,
+                                               cb, NULL);
+            }
+            if (r == NULL) {
+                Py_DECREF(ls);
+                return NULL;
+            }
+            Py_DECREF(r);
+        }
+        Py_DECREF(ls);
+    }
+    PyObject *ts = self->sc_timers;
+    self->sc_timers = NULL;
+    if (ts != NULL) {
+        Py_ssize_t n = PyList_GET_SIZE(ts);
+        for (Py_ssize_t i = 0; i < n; i++) {
+            PyObject *r = PyObject_CallMethodObjArgs(
+                PyList_GET_ITEM(ts, i), s_cancel, NULL);
+            if (r == NULL) {
+                Py_DECREF(ts);
+                return NULL;
+            }
+            Py_DECREF(r);
+        }
+        Py_DECREF(ts);
+    }
+    Py_RETURN_NONE;
+}
+
+PyObject *
+Scope_get_active(PyObject *self_, void *closure)
+{
+    (void)closure;
+    return PyBool_FromLong(((Scope *)self_)->sc_active);
+}
+
+int
+Scope_traverse(PyObject *self_, visitproc visit, void *arg)
+{
+    Scope *self = (Scope *)self_;
+    Py_VISIT(self->sc_fsm);
+    Py_VISIT(self->sc_listeners);
+    Py_VISIT(self->sc_timers);
+    return 0;
+}
+
+int
+Scope_clear_(PyObject *self_)
+{
+    Scope *self = (Scope *)self_;
+    Py_CLEAR(self->sc_fsm);
+    Py_CLEAR(self->sc_listeners);
+    Py_CLEAR(self->sc_timers);
+    return 0;
+}
+
+void
+Scope_dealloc(PyObject *self_)
+{
+    PyObject_GC_UnTrack(self_);
+    Scope_clear_(self_);
+    PyObject_GC_Del(self_);
+}
+
+PyMethodDef Scope_methods[] = {
+    {"on", (PyCFunction)(void (*)(void))Scope_on, METH_FASTCALL, NULL},
+    {"timeout", (PyCFunction)(void (*)(void))Scope_timeout, METH_FASTCALL,
+     NULL},
+    {"interval", (PyCFunction)(void (*)(void))Scope_interval,
+     METH_FASTCALL, NULL},
+    {"immediate", Scope_immediate, METH_O, NULL},
+    {"callback", Scope_callback, METH_O, NULL},
+    {"valid_transitions", Scope_valid_transitions, METH_O, NULL},
+    {"goto_state", Scope_goto_state, METH_O, NULL},
+    {"_dispose", Scope_dispose, METH_NOARGS, NULL},
+    {NULL, NULL, 0, NULL},
+};
+
+PyGetSetDef Scope_getset[] = {
+    {(char *)"active", Scope_get_active, NULL, NULL, NULL},
+    {NULL, NULL, NULL, NULL, NULL},
+};
+
+PyTypeObject ScopeType = {
+    PyVarObject_HEAD_INIT(NULL, 0)
+    "cueball_amd._speed.StateScope",
+    sizeof(Scope),
+    0,
+    Scope_dealloc,
+    0, 0, 0, 0, 0, 0, 0, 0, 0, 0, 0, 0, 0, 0,
+    Py_TPFLAGS_DEFAULT | Py_TPFLAGS_HAVE_GC,
+    "FSM state scope (native)",
+    Scope_traverse,
+    Scope_clear_,
+    0, 0, 0, 0,
+    Scope_methods,
+    0,
+    Scope_getset,
+};
+
+/* ------------------------------------------------------------------ */
+/* FSM                                                                 */
+/* ------------------------------------------------------------------ */
+
+struct FSMOb {
+    Emitter base;
+    PyObject *f_loop;
+    PyObject *f_state;          /* str or NULL */
+    PyObject *f_scope;          /* Scope* or NULL */
+    PyObject *f_valid;          /* sequence or NULL */
+    PyObject *f_pending;        /* str or NULL */
+    PyObject *f_emit_queue;     /* list */
+    PyObject *f_history;        /* list */
+    PyObject *f_flush_bound;    /* cached bound _flush_state_changed */
+    int f_entering;
+    int f_emit_scheduled;
+};
+
+extern PyTypeObject FSMType;
+
+PyObject *
+fsm_get_loop_of(FSMOb *fsm)
+{
+    return fsm->f_loop;
+}
+
+int
+fsm_flush_core(FSMOb *self)
+{
+    self->f_emit_scheduled = 0;
+    while (PyList_GET_SIZE(self->f_emit_queue) > 0) {
+        PyObject *st = PyList_GET_ITEM(self->f_emit_queue, 0);
+        Py_INCREF(st);
+        if (PyList_SetSlice(self->f_emit_queue, 0, 1, NULL) < 0) {
+            Py_DECREF(st);
+            return -1;
+        }
+        PyObject *eargs[2] = {s_stateChanged, st};
+        int r = emitter_emit_core(&self->base, eargs[0], eargs + 1, 1);
+        Py_DECREF(st);
+        if (r < 0)
+            return -1;
+    }
+    return 0;
+}
+
+PyObject *
+FSM_flush_state_changed(PyObject *self_, PyObject *noargs)
+{
+    (void)noargs;
+    if (fsm_flush_core((FSMOb *)self_) < 0)
+        return NULL;
+    Py_RETURN_NONE;
+}
+
+int
+fsm_queue_state_changed(FSMOb *self, PyObject *state)
+{
+    if (PyList_Append(self->f_emit_queue, state) < 0)
+        return -1;
+    if (!self->f_emit_scheduled) {
+        self->f_emit_scheduled = 1;
+        if (self->f_flush_bound == NULL) {
+            self->f_flush_bound = PyObject_GetAttr((PyObject *)self,
+                                                   s_flush_name);
+            if (self->f_flush_bound == NULL)
+                return -1;
+        }
+        PyObject *h = PyObject_CallMethodObjArgs(
+            self->f_loop, s_call_soon, self->f_flush_bound, NULL);
+        if (h == NULL)
+            return -1;
+        Py_DECREF(h);
+    }
+    return 0;
+}
+
+/* resolve "state_x_y" attr name for a state string, cached globally */
+PyObject *
+entry_attr_name(PyObject *state)
+{
+    PyObject *name = PyDict_GetItemWithError(g_entry_name_cache, state);
+    if (name != NULL) {
+        Py_INCREF(name);
+        return name;
+    }
+    if (PyErr_Occurred())
+        return NULL;
+    PyObject *replaced = PyObject_CallMethod(state, "replace", "OO",
+                                             s_dot, s_underscore);
+    if (replaced == NULL)
+        return NULL;
+    name = PyUnicode_Concat(s_state_prefix, replaced);
+    Py_DECREF(replaced);
+    if (name == NULL)
+        return NULL;
+    PyUnicode_InternInPlace(&name);
+    if (PyDict_SetItem(g_entry_name_cache, state, name) < 0) {
+        Py_DECREF(name);
+        return NULL;
+    }
+    return name;
+}
+
+int
+check_valid(FSMOb *self, PyObject *from, PyObject *state)
+{
+    if (self->f_valid == NULL)
+        return 0;
+    int c = PySequence_Contains(self->f_valid, state);
+    if (c < 0)
+        return -1;
+    if (!c) {
+        PyErr_Format(g_fsm_error,
+                     "%s: invalid transition %R -> %R (valid: %R)",
+                     Py_TYPE(self)->tp_name,
+                     from ? from : Py_None, state, self->f_valid);
+        return -1;
+    }
+    return 0;
+}
+
+int
+fsm_enter_loop(FSMOb *self, PyObject *state)
+{
+    PyObject *next_state = state;
+    Py_INCREF(next_state);
+    while (next_state != NULL) {
+        PyObject *target = next_state;
+        next_state = NULL;
+        if (self->f_scope != NULL) {
+            PyObject *old = self->f_scope;
+            self->f_scope = NULL;
+            PyObject *r = Scope_dispose(old, NULL);
+            Py_DECREF(old);
+            if (r == NULL) {
+                Py_DECREF(target);
+                return -1;
+            }
+            Py_DECREF(r);
+        }
+        Py_CLEAR(self->f_valid);
+        Py_INCREF(target);
+        Py_XSETREF(self->f_state, target);
+        if (PyList_Append(self->f_history, target) < 0) {
+            Py_DECREF(target);
+            return -1;
+        }
+        if (PyList_GET_SIZE(self->f_history) > 8) {
+            if (PyList_SetSlice(self->f_history, 0, 1, NULL) < 0) {
+                Py_DECREF(target);
+                return -1;
+            }
+        }
+        Scope *scope = PyObject_GC_New(Scope, &ScopeType);
+        if (scope == NULL) {
+            Py_DECREF(target);
+            return -1;
+        }
+        Py_INCREF((PyObject *)self);
+        scope->sc_fsm = (PyObject *)self;
+        scope->sc_listeners = NULL;
+        scope->sc_timers = NULL;
+        scope->sc_active = 1;
+        PyObject_GC_Track((PyObject *)scope);
+        Py_INCREF((PyObject *)scope);
+        Py_XSETREF(self->f_scope, (PyObject *)scope);
+
+        PyObject *attr = entry_attr_name(target);
+        if (attr == NULL) {
+            Py_DECREF(scope);
+            Py_DECREF(target);
+            return -1;
+        }
+        PyObject *entry = PyObject_GetAttr((PyObject *)self, attr);
+        Py_DECREF(attr);
+        if (entry == NULL) {
+            PyErr_Clear();
+            PyErr_Format(g_fsm_error,
+                         "%s has no state-entry function for %R",
+                         Py_TYPE(self)->tp_name, target);
+            Py_DECREF(scope);
+            Py_DECREF(target);
+            return -1;
+        }
+
+        self->f_entering = 1;
+        PyObject *r = PyObject_CallOneArg(entry, (PyObject *)scope);
+        Py_DECREF(entry);
+        Py_DECREF(scope);
+        self->f_entering = 0;
+        PyObject *pend = self->f_pending;
+        self->f_pending = NULL;
+        if (r == NULL) {
+            Py_XDECREF(pend);
+            Py_DECREF(target);
+            return -1;
+        }
+        Py_DECREF(r);
+        if (fsm_queue_state_changed(self, target) < 0) {
+            Py_XDECREF(pend);
+            Py_DECREF(target);
+            return -1;
+        }
+        if (pend != NULL) {
+            if (check_valid(self, target, pend) < 0) {
+                Py_DECREF(pend);
+                Py_DECREF(target);
+                return -1;
+            }
+            next_state = pend;
+        }
+        Py_DECREF(target);
+    }
+    return 0;
+}
+
+int
+fsm_goto_state(FSMOb *self, PyObject *state)
+{
+    if (check_valid(self, self->f_state, state) < 0)
+        return -1;
+    if (self->f_entering) {
+        if (self->f_pending != NULL) {
+            int eq = PyObject_RichCompareBool(self->f_pending, state, Py_EQ);
+            if (eq < 0)
+                return -1;
+            if (!eq) {
+                PyErr_Format(g_fsm_error,
+                             "%s: conflicting deferred transitions %R and "
+                             "%R from %R", Py_TYPE(self)->tp_name,
+                             self->f_pending, state,
+                             self->f_state ? self->f_state : Py_None);
+                return -1;
+            }
+            return 0;
+        }
+        Py_INCREF(state);
+        self->f_pending = state;
+        return 0;
+    }
+    return fsm_enter_loop(self, state);
+}
+
+int
+FSM_init(PyObject *self_, PyObject *args, PyObject *kwds)
+{
+    FSMOb *self = (FSMOb *)self_;
+    PyObject *initial = NULL, *loop = Py_None;
+    static const char *kwlist[] = {"initial_state", "loop", NULL};
+    if (!PyArg_ParseTupleAndKeywords(args, kwds, "U|O",
+                                     const_cast<char **>(kwlist),
+                                     &initial, &loop))
+        return -1;
+    if (Emitter_init(self_, NULL, NULL) < 0)
+        return -1;
+    PyObject *resolved = PyObject_CallOneArg(g_get_loop, loop);
+    if (resolved == NULL)
+        return -1;
+    Py_XSETREF(self->f_loop, resolved);
+    if (self->f_emit_queue == NULL) {
+        self->f_emit_queue = PyList_New(0);
+        if (self->f_emit_queue == NULL)
+            return -1;
+    }
+    if (self->f_history == NULL) {
+        self->f_history = PyList_New(0);
+        if (self->f_history == NULL)
+            return -1;
+    }
+    Py_CLEAR(self->f_state);
+    Py_CLEAR(self->f_valid);
+    Py_CLEAR(self->f_pending);
+    self->f_entering = 0;
+    return fsm_goto_state(self, initial);
+}
+
+PyObject *
+FSM_get_state(PyObject *self_, PyObject *noargs)
+{
+    FSMOb *self = (FSMOb *)self_;
+    (void)noargs;
+    if (self->f_state == NULL) {
+        PyErr_SetString(g_fsm_error, "FSM has no state yet");
+        return NULL;
+    }
+    Py_INCREF(self->f_state);
+    return self->f_state;
+}
+
+PyObject *
+FSM_is_in_state(PyObject *self_, PyObject *state)
+{
+    FSMOb *self = (FSMOb *)self_;
+    if (self->f_state == NULL)
+        Py_RETURN_FALSE;
+    int eq = PyUnicode_Compare(self->f_state, state);
+    if (eq == -1 && PyErr_Occurred())
+        return NULL;
+    if (eq == 0)
+        Py_RETURN_TRUE;
+    /* prefix semantics: cur startswith state + "." */
+    Py_ssize_t sl = PyUnicode_GET_LENGTH(state);
+    Py_ssize_t cl = PyUnicode_GET_LENGTH(self->f_state);
+    if (cl > sl) {
+        int m = PyUnicode_Tailmatch(self->f_state, state, 0, sl, -1);
+        if (m < 0)
+            return NULL;
+        if (m && PyUnicode_ReadChar(self->f_state, sl) == '.')
+            Py_RETURN_TRUE;
+    }
+    Py_RETURN_FALSE;
+}
+
+PyObject *
+FSM_get_state_history(PyObject *self_, PyObject *noargs)
+{
+    FSMOb *self = (FSMOb *)self_;
+    (void)noargs;
+    return PyList_GetSlice(self->f_history, 0,
+                           PyList_GET_SIZE(self->f_history));
+}
+
+PyObject *
+FSM_goto_state_py(PyObject *self_, PyObject *state)
+{
+    if (!PyUnicode_Check(state)) {
+        PyErr_SetString(PyExc_TypeError, "state must be a str");
+        return NULL;
+    }
+    if (fsm_goto_state((FSMOb *)self_, state) < 0)
+        return NULL;
+    Py_RETURN_NONE;
+}
+
+PyObject *
+FSM_get__fsm_state(PyObject *self_, void *closure)
+{
+    FSMOb *self = (FSMOb *)self_;
+    (void)closure;
+    if (self->f_state == NULL)
+        Py_RETURN_NONE;
+    Py_INCREF(self->f_state);
+    return self->f_state;
+}
+
+PyObject *
+FSM_get__fsm_valid(PyObject *self_, void *closure)
+{
+    FSMOb *self = (FSMOb *)self_;
+    (void)closure;
+    if (self->f_valid == NULL)
+        Py_RETURN_NONE;
+    Py_INCREF(self->f_valid);
+    return self->f_valid;
+}
+
+int
+FSM_set__fsm_valid(PyObject *self_, PyObject *value, void *closure)
+{
+    FSMOb *self = (FSMOb *)self_;
+    (void)closure;
+    if (value == NULL || value == Py_None) {
+        Py_CLEAR(self->f_valid);
+        return 0;
+    }
+    Py_INCREF(value);
+    Py_XSETREF(self->f_valid, value);
+    return 0;
+}
+
+PyObject *
+Scope_valid_transitions(PyObject *self_, PyObject *states)
+{
+    Scope *self = (Scope *)self_;
+    FSMOb *fsm = (FSMOb *)self->sc_fsm;
+    Py_INCREF(states);
+    Py_XSETREF(fsm->f_valid, states);
+    Py_RETURN_NONE;
+}
+
+/* S.interval implemented natively with a tiny repeating driver */
+typedef struct {
+    PyObject_HEAD
+    PyObject *iv_scope;
+    PyObject *iv_cb;
+    PyObject *iv_secs;
+    PyObject *iv_handle;   /* current timer handle */
+    int iv_stopped;
+} IntervalOb;
+
+extern PyTypeObject IntervalType;
+
+PyObject *
+Interval_call(PyObject *self_, PyObject *args, PyObject *kwds)
+{
+    IntervalOb *self = (IntervalOb *)self_;
+    (void)args; (void)kwds;
+    if (self->iv_stopped || !((Scope *)self->iv_scope)->sc_active)
+        Py_RETURN_NONE;
+    PyObject *r = PyObject_CallNoArgs(self->iv_cb);
+    if (r == NULL)
+        return NULL;
+    Py_DECREF(r);
+    if (!self->iv_stopped && ((Scope *)self->iv_scope)->sc_active) {
+        FSMOb *fsm = (FSMOb *)((Scope *)self->iv_scope)->sc_fsm;
+        PyObject *h = PyObject_CallMethodObjArgs(
+            fsm->f_loop, s_call_later, self->iv_secs, self_, NULL);
+        if (h == NULL)
+            return NULL;
+        Py_XSETREF(self->iv_handle, h);
+    }
+    Py_RETURN_NONE;
+}
+
+PyObject *
+Interval_cancel(PyObject *self_, PyObject *noargs)
+{
+    IntervalOb *self = (IntervalOb *)self_;
+    (void)noargs;
+    self->iv_stopped = 1;
+    if (self->iv_handle != NULL) {
+        PyObject *r = PyObject_CallMethodObjArgs(self->iv_handle, s_cancel,
+                                                 NULL);
+        if (r == NULL)
+            return NULL;
+        Py_DECREF(r);
+    }
+    Py_RETURN_NONE;
+}
+
+int
+Interval_traverse(PyObject *self_, visitproc visit, void *arg)
+{
+    IntervalOb *self = (IntervalOb *)self_;
+    Py_VISIT(self->iv_scope);
+    Py_VISIT(self->iv_cb);
+    Py_VISIT(self->iv_handle);
+    return 0;
+}
+
+int
+Interval_clear_(PyObject *self_)
+{
+    IntervalOb *self = (IntervalOb *)self_;
+    Py_CLEAR(self->iv_scope);
+    Py_CLEAR(self->iv_cb);
+    Py_CLEAR(self->iv_secs);
+    Py_CLEAR(self->iv_handle);
+    return 0;
+}
+
+void
+Interval_dealloc(PyObject *self_)
+{
+    PyObject_GC_UnTrack(self_);
+    Interval_clear_(self_);
+    PyObject_GC_Del(self_);
+}
+
+PyMethodDef Interval_methods[] = {
+    {"cancel", Interval_cancel, METH_NOARGS, NULL},
+    {NULL, NULL, 0, NULL},
+};
+
+PyTypeObject IntervalType = {
+    PyVarObject_HEAD_INIT(NULL, 0)
+    "cueball_amd._speed._Interval",
+    sizeof(IntervalOb),
+    0,
+    Interval_dealloc,
+    0, 0, 0, 0, 0, 0, 0, 0, 0,
+    Interval_call,
+    0, 0, 0, 0,
+    Py_TPFLAGS_DEFAULT | Py_TPFLAGS_HAVE_GC,
+    0,
+    Interval_traverse,
+    Interval_clear_,
+    0, 0, 0, 0,
+    Interval_methods,
+};
+
+PyObject *
+Scope_interval(PyObject *self_, PyObject *const *args, Py_ssize_t nargs)
+{
+    Scope *self = (Scope *)self_;
+    if (nargs != 2) {
+        PyErr_SetString(PyExc_TypeError, "S.interval(ms, cb)");
+        return NULL;
+    }
+    if (scope_check_active(self, "S.interval()") < 0)
+        return NULL;
+    double ms = PyFloat_AsDouble(args[0]);
+    if (ms == -1.0 && PyErr_Occurred())
+        return NULL;
+    IntervalOb *iv = PyObject_GC_New(IntervalOb, &IntervalType);
+    if (iv == NULL)
+        return NULL;
+    Py_INCREF(self_);
+    iv->iv_scope = self_;
+    Py_INCREF(args[1]);
+    iv->iv_cb = args[1];
+    iv->iv_secs = PyFloat_FromDouble(ms / 1000.0);
+    iv->iv_handle = NULL;
+    iv->iv_stopped = 0;
+    PyObject_GC_Track((PyObject *)iv);
+    if (iv->iv_secs == NULL) {
+        Py_DECREF(iv);
+        return NULL;
+    }
+    FSMOb *fsm = (FSMOb *)self->sc_fsm;
+    PyObject *h = PyObject_CallMethodObjArgs(fsm->f_loop, s_call_later,
+                                             iv->iv_secs, (PyObject *)iv,
+                                             NULL);
+    if (h == NULL) {
+        Py_DECREF(iv);
+        return NULL;
+    }
+    iv->iv_handle = h;
+    int rc = scope_add_timer(self, (PyObject *)iv);
+    Py_DECREF(iv);
+    if (rc < 0)
+        return NULL;
+    Py_RETURN_NONE;
+}
+
+int
+FSM_traverse(PyObject *self_, visitproc visit, void *arg)
+{
+    FSMOb *self = (FSMOb *)self_;
+    Py_VISIT(self->f_loop);
+    Py_VISIT(self->f_state);
+    Py_VISIT(self->f_scope);
+    Py_VISIT(self->f_valid);
+    Py_VISIT(self->f_pending);
+    Py_VISIT(self->f_emit_queue);
+    Py_VISIT(self->f_history);
+    Py_VISIT(self->f_flush_bound);
+    return Emitter_traverse(self_, visit, arg);
+}
+
+int
+FSM_clear_(PyObject *self_)
+{
+    FSMOb *self = (FSMOb *)self_;
+    Py_CLEAR(self->f_loop);
+    Py_CLEAR(self->f_state);
+    Py_CLEAR(self->f_scope);
+    Py_CLEAR(self->f_valid);
+    Py_CLEAR(self->f_pending);
+    Py_CLEAR(self->f_emit_queue);
+    Py_CLEAR(self->f_history);
+    Py_CLEAR(self->f_flush_bound);
+    return Emitter_clear_(self_);
+}
+
+void
+FSM_dealloc(PyObject *self_)
+{
+    FSMOb *self = (FSMOb *)self_;
+    PyTypeObject *tp = Py_TYPE(self_);
+    PyObject_GC_UnTrack(self_);
+    if (self->base.ev_weakrefs != NULL)
+        PyObject_ClearWeakRefs(self_);
+    FSM_clear_(self_);
+    tp->tp_free(self_);
+}
+
+PyMethodDef FSM_methods[] = {
+    {"get_state", FSM_get_state, METH_NOARGS, NULL},
+    {"is_in_state", FSM_is_in_state, METH_O, NULL},
+    {"get_state_history", FSM_get_state_history, METH_NOARGS, NULL},
+    {"goto_state", FSM_goto_state_py, METH_O, NULL},
+    {"_flush_state_changed", FSM_flush_state_changed, METH_NOARGS, NULL},
+    {NULL, NULL, 0, NULL},
+};
+
+PyMemberDef FSM_members[] = {
+    {(char *)"_loop", T_OBJECT, offsetof(FSMOb, f_loop), READONLY, NULL},
+    {NULL, 0, 0, 0, NULL},
+};
+
+PyGetSetDef FSM_getset[] = {
+    {(char *)"_fsm_state", FSM_get__fsm_state, NULL, NULL, NULL},
+    {(char *)"_fsm_valid", FSM_get__fsm_valid, FSM_set__fsm_valid, NULL,
+     NULL},
+    {NULL, NULL, NULL, NULL, NULL},
+};
+
+PyTypeObject FSMType = {
+    PyVarObject_HEAD_INIT(NULL, 0)
+    "cueball_amd._speed.FSM",
+    sizeof(FSMOb),
+    0,
+    FSM_dealloc,
+    0, 0, 0, 0, 0, 0, 0, 0, 0, 0, 0, 0, 0, 0,
+    Py_TPFLAGS_DEFAULT | Py_TPFLAGS_BASETYPE | Py_TPFLAGS_HAVE_GC,
+    "Moore machine (native)",
+    FSM_traverse,
+    FSM_clear_,
+    0, 0, 0, 0,
+    FSM_methods,
+    FSM_members,
+    FSM_getset,
+    &EmitterType,                         /* tp_base */
+    0, 0, 0,
+    0,                                    /* tp_dictoffset (inherited) */
+    FSM_init,
+    0,
+    PyType_GenericNew,
+};
+
+/* ------------------------------------------------------------------ */
+/* module                                                              */
+/* ------------------------------------------------------------------ */
+
+PyObject *
+speed_set_helpers(PyObject *mod, PyObject *const *args, Py_ssize_t nargs)
+{
+    (void)mod;
+    if (nargs != 2) {
+        PyErr_SetString(PyExc_TypeError, "_set_helpers(get_loop, FSMError)");
+        return NULL;
+    }
+    Py_INCREF(args[0]);
+    Py_XSETREF(g_get_loop, args[0]);
+    Py_INCREF(args[1]);
+    Py_XSETREF(g_fsm_error, args[1]);
+    Py_RETURN_NONE;
+}
+
+PyMethodDef speed_methods[] = {
+    {"_set_helpers", (PyCFunction)(void (*)(void))speed_set_helpers,
+     METH_FASTCALL, NULL},
+    {NULL, NULL, 0, NULL},
+};
+
+struct PyModuleDef speedmodule = {
+    PyModuleDef_HEAD_INIT,
+    "cueball_amd._speed",
+    "Native event/FSM runtime core",
+    -1,
+    speed_methods,
+};
+
+}  /* namespace */
+
+PyMODINIT_FUNC
+PyInit__speed(void)
+{
+    s_stateChanged = PyUnicode_InternFromString("stateChanged");
+    s_listener = PyUnicode_InternFromString("listener");
+    s_on = PyUnicode_InternFromString("on");
+    s_remove_listener = PyUnicode_InternFromString("remove_listener");
+    s_call_soon = PyUnicode_InternFromString("call_soon");
+    s_call_later = PyUnicode_InternFromString("call_later");
+    s_cancel = PyUnicode_InternFromString("cancel");
+    s_state_prefix = PyUnicode_InternFromString("state_");
+    s_dot = PyUnicode_InternFromString(".");
+    s_underscore = PyUnicode_InternFromString("_");
+    s_flush_name = PyUnicode_InternFromString("_flush_state_changed");
+    g_entry_name_cache = PyDict_New();
+    if (g_entry_name_cache == NULL)
+        return NULL;
+
+    if (PyType_Ready(&EmitterType) < 0 ||
+        PyType_Ready(&OnceWrapperType) < 0 ||
+        PyType_Ready(&GuardedCbType) < 0 ||
+        PyType_Ready(&ScopeType) < 0 ||
+        PyType_Ready(&IntervalType) < 0 ||
+        PyType_Ready(&FSMType) < 0)
+        return NULL;
+
+    PyObject *m = PyModule_Create(&speedmodule);
+    if (m == NULL)
+        return NULL;
+    Py_INCREF(&EmitterType);
+    PyModule_AddObject(m, "EventEmitter", (PyObject *)&EmitterType);
+    Py_INCREF(&ScopeType);
+    PyModule_AddObject(m, "StateScope", (PyObject *)&ScopeType);
+    Py_INCREF(&FSMType);
+    PyModule_AddObject(m, "FSM", (PyObject *)&FSMType);
+    return m;
+}

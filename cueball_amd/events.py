@@ -84,3 +84,20 @@ class EventEmitter:
         for listener in tuple(ls):
             listener(*args)
         return True
+
+
+# ---------------------------------------------------------------------------
+# Native core: when the C++ extension is built (cueball_amd._speed), its
+# EventEmitter replaces the pure-Python one above — identical semantics
+# (the full test suite runs against either; set CUEBALL_PURE=1 to force
+# the Python implementation).
+import os as _os
+
+PurePythonEventEmitter = EventEmitter
+NATIVE = False
+if not _os.environ.get("CUEBALL_PURE"):
+    try:
+        from ._speed import EventEmitter  # noqa: F811
+        NATIVE = True
+    except ImportError:
+        pass

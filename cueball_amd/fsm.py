@@ -322,3 +322,23 @@ class FSM(EventEmitter):
         while q:
             st = q.pop(0)
             self.emit("stateChanged", st)
+
+
+# ---------------------------------------------------------------------------
+# Native core: when cueball_amd._speed is built, its FSM/StateScope
+# replace the pure-Python ones above (same semantics, validated by the
+# same test suite; CUEBALL_PURE=1 forces the Python implementation).
+import os as _os
+
+PurePythonFSM = FSM
+PurePythonStateScope = StateScope
+NATIVE = False
+if not _os.environ.get("CUEBALL_PURE"):
+    try:
+        from . import _speed as _speed_mod
+        _speed_mod._set_helpers(get_loop, FSMError)
+        FSM = _speed_mod.FSM  # noqa: F811
+        StateScope = _speed_mod.StateScope  # noqa: F811
+        NATIVE = True
+    except ImportError:
+        pass

@@ -1,0 +1,34 @@
+"""Build the cueball_amd package and its native speed core.
+
+The native extension is plain C++ (no GPU code exists in this problem
+domain — the reference is a pure connection-pool library);
+it accelerates the event/FSM hot path that every claim/release walks.
+Built in-tree (build_ext --inplace) so the .so ships with the source.
+"""
+
+import os
+
+from setuptools import Extension, setup
+
+ROOT = os.path.dirname(os.path.abspath(__file__))
+
+ext_modules = []
+speed_src = os.path.join("cueball_amd", "_native", "speed.cpp")
+if os.path.exists(os.path.join(ROOT, speed_src)):
+    ext_modules.append(Extension(
+        "cueball_amd._speed",
+        sources=[speed_src],
+        language="c++",
+        extra_compile_args=["-O3", "-std=c++17", "-fvisibility=hidden"],
+    ))
+
+setup(
+    name="cueball-amd",
+    version="0.1.0",
+    description="Connection pooling + DNS service discovery framework "
+                "(node-cueball capabilities, asyncio + C++ core)",
+    packages=["cueball_amd"],
+    ext_modules=ext_modules,
+    python_requires=">=3.8",
+    scripts=["bin/cbresolve"],
+)
