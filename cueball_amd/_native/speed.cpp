@@ -36,10 +36,13 @@ PyObject *s_state_prefix;      /* "state_" */
 PyObject *s_dot;               /* "." */
 PyObject *s_underscore;        /* "_" */
 PyObject *s_flush_name;        /* "_flush_state_changed" */
+PyObject *s_internal;          /* "_cueball_internal" */
+PyObject *s_is_closed;         /* "is_closed" */
 
 PyObject *g_get_loop;          /* python helper: get_loop(loop) */
 PyObject *g_fsm_error;         /* exception class FSMError */
 PyObject *g_entry_name_cache;  /* dict: state name -> "state_x_y" */
+PyObject *g_flush_batches;     /* dict: loop -> _FlushBatch */
 
 /* ------------------------------------------------------------------ */
 /* EventEmitter                                                        */
@@ -863,25 +866,124 @@ FSM_flush_state_changed(PyObject *self_, PyObject *noargs)
     Py_RETURN_NONE;
 }
 
+/* Batched flush: all FSMs that queue a stateChanged in the same loop
+ * turn share one call_soon.  Relative order is preserved (FSMs drain in
+ * first-queue order; each FSM drains its whole queue, exactly as the
+ * per-FSM call_soon did).  A batch already being drained is never
+ * appended to: later queuers start a fresh batch + call_soon, keeping
+ * the ordering identical to the unbatched scheme. */
+typedef struct {
+    PyObject_HEAD
+    PyObject *fb_loop;
+    PyObject *fb_list;   /* list of FSMOb */
+} FlushBatch;
+
+extern PyTypeObject FlushBatchType;
+
+PyObject *
+FlushBatch_call(PyObject *self_, PyObject *args, PyObject *kwds)
+{
+    FlushBatch *self = (FlushBatch *)self_;
+    (void)args; (void)kwds;
+    /* detach ourselves: new queuers start a new batch */
+    PyObject *cur = PyDict_GetItemWithError(g_flush_batches, self->fb_loop);
+    if (cur == (PyObject *)self) {
+        if (PyDict_DelItem(g_flush_batches, self->fb_loop) < 0)
+            return NULL;
+    } else if (cur == NULL && PyErr_Occurred()) {
+        return NULL;
+    }
+    Py_ssize_t n = PyList_GET_SIZE(self->fb_list);
+    for (Py_ssize_t i = 0; i < n; i++) {
+        FSMOb *fsm = (FSMOb *)PyList_GET_ITEM(self->fb_list, i);
+        if (fsm_flush_core(fsm) < 0)
+            return NULL;
+    }
+    Py_RETURN_NONE;
+}
+
+int
+FlushBatch_traverse(PyObject *self_, visitproc visit, void *arg)
+{
+    FlushBatch *self = (FlushBatch *)self_;
+    Py_VISIT(self->fb_loop);
+    Py_VISIT(self->fb_list);
+    return 0;
+}
+
+int
+FlushBatch_clear_(PyObject *self_)
+{
+    FlushBatch *self = (FlushBatch *)self_;
+    Py_CLEAR(self->fb_loop);
+    Py_CLEAR(self->fb_list);
+    return 0;
+}
+
+void
+FlushBatch_dealloc(PyObject *self_)
+{
+    PyObject_GC_UnTrack(self_);
+    FlushBatch_clear_(self_);
+    PyObject_GC_Del(self_);
+}
+
+PyTypeObject FlushBatchType = {
+    PyVarObject_HEAD_INIT(NULL, 0)
+    "cueball_amd._speed._FlushBatch",
+    sizeof(FlushBatch),
+    0,
+    FlushBatch_dealloc,
+    0, 0, 0, 0, 0, 0, 0, 0, 0,
+    FlushBatch_call,
+    0, 0, 0, 0,
+    Py_TPFLAGS_DEFAULT | Py_TPFLAGS_HAVE_GC,
+    0,
+    FlushBatch_traverse,
+    FlushBatch_clear_,
+};
+
 int
 fsm_queue_state_changed(FSMOb *self, PyObject *state)
 {
     if (PyList_Append(self->f_emit_queue, state) < 0)
         return -1;
-    if (!self->f_emit_scheduled) {
-        self->f_emit_scheduled = 1;
-        if (self->f_flush_bound == NULL) {
-            self->f_flush_bound = PyObject_GetAttr((PyObject *)self,
-                                                   s_flush_name);
-            if (self->f_flush_bound == NULL)
-                return -1;
+    if (self->f_emit_scheduled)
+        return 0;
+    self->f_emit_scheduled = 1;
+    PyObject *batch = PyDict_GetItemWithError(g_flush_batches, self->f_loop);
+    if (batch == NULL) {
+        if (PyErr_Occurred())
+            return -1;
+        FlushBatch *fb = PyObject_GC_New(FlushBatch, &FlushBatchType);
+        if (fb == NULL)
+            return -1;
+        Py_INCREF(self->f_loop);
+        fb->fb_loop = self->f_loop;
+        fb->fb_list = PyList_New(0);
+        PyObject_GC_Track((PyObject *)fb);
+        if (fb->fb_list == NULL) {
+            Py_DECREF(fb);
+            return -1;
+        }
+        if (PyDict_SetItem(g_flush_batches, self->f_loop,
+                           (PyObject *)fb) < 0) {
+            Py_DECREF(fb);
+            return -1;
         }
         PyObject *h = PyObject_CallMethodObjArgs(
-            self->f_loop, s_call_soon, self->f_flush_bound, NULL);
-        if (h == NULL)
+            self->f_loop, s_call_soon, (PyObject *)fb, NULL);
+        if (h == NULL) {
+            Py_DECREF(fb);
             return -1;
+        }
         Py_DECREF(h);
+        batch = (PyObject *)fb;
+        Py_DECREF(fb);  /* dict holds it */
     }
+    if (PyList_Append(((FlushBatch *)batch)->fb_list,
+                      (PyObject *)self) < 0)
+        return -1;
     return 0;
 }
 
@@ -1439,7 +1541,78 @@ speed_set_helpers(PyObject *mod, PyObject *const *args, Py_ssize_t nargs)
     Py_RETURN_NONE;
 }
 
+PyObject *
+speed_count_listeners(PyObject *mod, PyObject *const *args,
+                      Py_ssize_t nargs)
+{
+    (void)mod;
+    if (nargs != 2) {
+        PyErr_SetString(PyExc_TypeError, "count_listeners(emitter, event)");
+        return NULL;
+    }
+    if (!PyObject_TypeCheck(args[0], &EmitterType)) {
+        PyErr_SetString(PyExc_TypeError, "emitter must be an EventEmitter");
+        return NULL;
+    }
+    Emitter *em = (Emitter *)args[0];
+    if (em->ev_events == NULL)
+        return PyLong_FromLong(0);
+    PyObject *ls = PyDict_GetItemWithError(em->ev_events, args[1]);
+    if (ls == NULL)
+        return PyErr_Occurred() ? NULL : PyLong_FromLong(0);
+    Py_ssize_t n = PyList_GET_SIZE(ls);
+    long count = 0;
+    for (Py_ssize_t i = 0; i < n; i++) {
+        PyObject *h = PyList_GET_ITEM(ls, i);
+        if (!PyCallable_Check(h))
+            continue;
+        PyObject *marker = NULL;
+        if (_PyObject_LookupAttr(h, s_internal, &marker) < 0)
+            return NULL;
+        if (marker != NULL) {
+            int truthy = PyObject_IsTrue(marker);
+            Py_DECREF(marker);
+            if (truthy < 0)
+                return NULL;
+            if (truthy)
+                continue;
+        }
+        /* once() wrappers: look at the original listener */
+        PyObject *target = NULL;
+        if (Py_TYPE(h) == &OnceWrapperType) {
+            target = ((OnceWrapper *)h)->ow_listener;
+            Py_INCREF(target);
+        } else if (_PyObject_LookupAttr(h, s_listener, &target) < 0) {
+            return NULL;
+        }
+        if (target != NULL && target != h) {
+            if (_PyObject_LookupAttr(target, s_internal, &marker) < 0) {
+                Py_DECREF(target);
+                return NULL;
+            }
+            if (marker != NULL) {
+                int truthy = PyObject_IsTrue(marker);
+                Py_DECREF(marker);
+                if (truthy < 0) {
+                    Py_DECREF(target);
+                    return NULL;
+                }
+                if (truthy) {
+                    Py_DECREF(target);
+                    continue;
+                }
+            }
+        }
+        Py_XDECREF(target);
+        count++;
+    }
+    return PyLong_FromLong(count);
+}
+
 PyMethodDef speed_methods[] = {
+    {"count_listeners",
+     (PyCFunction)(void (*)(void))speed_count_listeners, METH_FASTCALL,
+     NULL},
     {"_set_helpers", (PyCFunction)(void (*)(void))speed_set_helpers,
      METH_FASTCALL, NULL},
     {NULL, NULL, 0, NULL},
@@ -1469,8 +1642,13 @@ PyInit__speed(void)
     s_dot = PyUnicode_InternFromString(".");
     s_underscore = PyUnicode_InternFromString("_");
     s_flush_name = PyUnicode_InternFromString("_flush_state_changed");
+    s_internal = PyUnicode_InternFromString("_cueball_internal");
+    s_is_closed = PyUnicode_InternFromString("is_closed");
     g_entry_name_cache = PyDict_New();
     if (g_entry_name_cache == NULL)
+        return NULL;
+    g_flush_batches = PyDict_New();
+    if (g_flush_batches == NULL)
         return NULL;
 
     if (PyType_Ready(&EmitterType) < 0 ||
@@ -1478,6 +1656,7 @@ PyInit__speed(void)
         PyType_Ready(&GuardedCbType) < 0 ||
         PyType_Ready(&ScopeType) < 0 ||
         PyType_Ready(&IntervalType) < 0 ||
+        PyType_Ready(&FlushBatchType) < 0 ||
         PyType_Ready(&FSMType) < 0)
         return NULL;
 

@@ -44,6 +44,15 @@ from .logutil import CueballLogger
 
 __all__ = ["SocketMgrFSM", "ClaimHandle", "ConnectionSlotFSM"]
 
+try:
+    import os as _os
+    if _os.environ.get("CUEBALL_PURE"):
+        _native_count = None
+    else:
+        from ._speed import count_listeners as _native_count
+except ImportError:
+    _native_count = None
+
 
 def _call_optional(obj: Any, *names: str) -> None:
     for name in names:
@@ -494,6 +503,8 @@ class ClaimHandle(FSM):
 def count_listeners(emitter: Any, event: str) -> int:
     """Count user-registered listeners, ignoring cueball's own internal
     handlers (lib/connection-fsm.js:786-808)."""
+    if _native_count is not None and isinstance(emitter, EventEmitter):
+        return _native_count(emitter, event)
     # fast path for our own EventEmitter: read the list in place
     ev = getattr(emitter, "_events", None)
     if ev is not None:

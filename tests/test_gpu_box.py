@@ -21,12 +21,15 @@ ROOT = os.path.dirname(os.path.dirname(os.path.abspath(__file__)))
 
 
 def test_native_extension_loads():
-    """The native claim-path extension must be importable on the box —
-    no silent pure-Python fallback."""
-    try:
-        from cueball_amd import _speed  # noqa: F401
-    except ImportError:
-        pytest.skip("native extension not yet built in this round")
+    """The native runtime core must be loaded on the box — no silent
+    pure-Python fallback (the .so is built in-tree and ships with the
+    snapshot)."""
+    from cueball_amd import _speed  # noqa: F401
+    import cueball_amd.fsm as fsm
+    import cueball_amd.events as events
+    assert fsm.NATIVE, "FSM core fell back to pure Python"
+    assert events.NATIVE, "EventEmitter fell back to pure Python"
+    assert fsm.FSM is _speed.FSM
 
 
 def test_smoke_entrypoint():
