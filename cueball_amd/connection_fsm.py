@@ -416,8 +416,12 @@ class ClaimHandle(FSM):
     def _relinquish(self, state: str) -> None:
         if self._fsm_state != "claimed":
             if self.is_in_state("released") or self.is_in_state("closed"):
-                stack = self.ch_release_stack or ["?", "?", "?"]
-                by = stack[2] if len(stack) > 2 else stack[-1]
+                stack = self.ch_release_stack or ["?"]
+                # python stacks are innermost-last: the releasing call
+                # site sits 2 frames above the capture (release ->
+                # _relinquish); node's stack[2] is the same frame
+                # counted from the other end (lib/connection-fsm.js:603)
+                by = stack[-3] if len(stack) >= 3 else stack[-1]
                 raise mod_errors.CueballError(
                     "Connection not claimed by this handle, released by %s"
                     % by)

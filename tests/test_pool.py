@@ -599,3 +599,228 @@ def test_claim_async_sugar():
         hdl.release()
 
     run_vt(lambda loop: body(loop))
+
+
+def test_cueball_108_close_then_conn_close():
+    """hdl.close() racing the conn's own 'close' event must not wedge
+    the pool or double-count the slot (reference #108)."""
+    async def body(loop):
+        recovery = {"default": {"timeout": 500, "retries": 2, "delay": 0}}
+        ctx = Ctx(loop, spares=2, maximum=2, recovery=recovery)
+        ctx.add_backend("b1")
+        await settle(loop)
+        for c in list(ctx.connections):
+            c.connect()
+        await advance(loop, 0.1)
+        assert ctx.pool.is_in_state("running")
+        assert len(ctx.connections) == 2
+
+        _, box = ctx.claim()
+        await settle(loop)
+        assert box["err"] is None
+        await advance(loop, 0.1)
+        box["hdl"].close()
+        box["conn"].emit("close")
+        await advance(loop, 0.1)
+        assert ctx.pool.is_in_state("running")
+
+        ctx.pool.stop()
+        await advance(loop, 2.0)
+        assert ctx.pool.is_in_state("stopped")
+
+    run_vt(lambda loop: body(loop))
+
+
+def test_cueball_111_close_then_conn_error():
+    """hdl.close() racing a conn 'error' (reference #111)."""
+    async def body(loop):
+        recovery = {"default": {"timeout": 500, "retries": 2, "delay": 0}}
+        ctx = Ctx(loop, spares=2, maximum=2, recovery=recovery)
+        ctx.add_backend("b1")
+        await settle(loop)
+        for c in list(ctx.connections):
+            c.connect()
+        await advance(loop, 0.1)
+        _, box = ctx.claim()
+        await settle(loop)
+        assert box["err"] is None
+        await advance(loop, 0.1)
+        box["hdl"].close()
+        box["conn"].emit("error", RuntimeError("Foo"))
+        await advance(loop, 0.1)
+        assert ctx.pool.is_in_state("running")
+
+        ctx.pool.stop()
+        await advance(loop, 2.0)
+        assert ctx.pool.is_in_state("stopped")
+
+    run_vt(lambda loop: body(loop))
+
+
+def test_backend_failure_removal_race_144():
+    """A backend removed from the resolver while its slots are failing
+    must not leave the pool counting it dead (reference #144)."""
+    async def body(loop):
+        recovery = {"default": {"timeout": 500, "retries": 2, "delay": 0}}
+        ctx = Ctx(loop, spares=2, maximum=2, recovery=recovery)
+        ctx.add_backend("b1")
+        ctx.add_backend("b2")
+        await settle(loop)
+        assert ctx.counts() == {"b1": 1, "b2": 1}
+        ctx.by_backend("b1")[0].connect()
+        ctx.by_backend("b2")[0].connect()
+        await advance(loop, 0.1)
+
+        conns = list(ctx.connections)
+        for c in conns:
+            c.emit("error", RuntimeError("test"))
+        await advance(loop, 0.1)
+        assert ctx.pool.is_in_state("running")
+        assert ctx.pool.get_last_error() is None
+
+        ctx.resolver.remove("b2")
+        for c in list(ctx.connections):
+            c.emit("error", RuntimeError("test2"))
+        await advance(loop, 0.1)
+        assert ctx.pool.is_in_state("failed")
+        assert ctx.pool.p_keys == ["b1"]
+        assert ctx.pool.p_dead == {"b1": True}
+
+        ctx.pool.stop()
+        # the monitor slot needs a full connect-timeout cycle to notice
+        # it is unwanted
+        await advance(loop, 8.0)
+        assert ctx.pool.is_in_state("stopped")
+
+    run_vt(lambda loop: body(loop))
+
+
+def test_get_stats_shape():
+    async def body(loop):
+        ctx = Ctx(loop, spares=2, maximum=2)
+        s = ctx.pool.get_stats()
+        assert set(s.keys()) == {"counters", "totalConnections",
+                                 "idleConnections", "pendingConnections",
+                                 "waiterCount"}
+        assert s["totalConnections"] == 0
+
+    run_vt(lambda loop: body(loop))
+
+
+def test_pool_with_churn_rate_limit():
+    """maxChurnRate=4/s: the pool adds at most one connection per 250ms
+    per backend, and drains the same way (test/pool.test.js:1041)."""
+    async def body(loop):
+        ctx = Ctx(loop, spares=4, maximum=4, maxChurnRate=4.0)
+        ctx.add_backend("b1")
+        await settle(loop)
+        assert ctx.counts() == {"b1": 1}
+        ctx.connections[0].connect()
+
+        await advance(loop, 0.35)
+        assert ctx.counts() == {"b1": 2}
+        ctx.connections[1].connect()
+
+        await advance(loop, 0.25)
+        assert ctx.counts() == {"b1": 3}
+        ctx.connections[2].connect()
+
+        await advance(loop, 0.25)
+        assert ctx.counts() == {"b1": 4}
+        ctx.connections[3].connect()
+
+        ctx.add_backend("b2")
+        await advance(loop, 0.05)
+        # still rate-limited: no instant rebalance to b2
+        assert ctx.counts() == {"b1": 4}
+
+        await advance(loop, 0.2)
+        c = ctx.counts()
+        assert c.get("b2", 0) >= 1  # first b2 conn allowed now
+
+        ctx.pool.stop()
+        await advance(loop, 2.0)
+        assert ctx.pool.is_in_state("stopped")
+
+    run_vt(lambda loop: body(loop))
+
+
+def test_decoherence_shuffle_fires():
+    """The >=60s decoherence timer reshuffles the preference list and
+    rebalances (lib/pool.js:501-519) — observable on the virtual clock."""
+    async def body(loop):
+        import random
+        ctx = Ctx(loop, spares=2, maximum=4)
+        for b in ("b1", "b2", "b3", "b4"):
+            ctx.add_backend(b)
+        await settle(loop)
+        for c in list(ctx.connections):
+            c.connect()
+        await settle(loop)
+
+        seen_orders = {tuple(ctx.pool.p_keys)}
+        random.seed(1234)
+        for _ in range(5):
+            await advance(loop, 61)
+            for c in list(ctx.connections):
+                if not c.connected and not c.dead:
+                    c.connect()
+            seen_orders.add(tuple(ctx.pool.p_keys))
+        # the preference list must actually have moved at least once
+        assert len(seen_orders) > 1
+
+        ctx.pool.stop()
+        await advance(loop, 2.0)
+
+    run_vt(lambda loop: body(loop))
+
+
+def test_lpf_anti_shrink():
+    """Under sustained claim load, releasing everything at once must not
+    let the pool shrink immediately: the 128-tap EMA clamp holds the
+    target up (lib/pool.js:37-100, :579-588)."""
+    async def body(loop):
+        ctx = Ctx(loop, spares=1, maximum=8)
+        ctx.add_backend("b1")
+        ctx.add_backend("b2")
+        await settle(loop)
+        for c in list(ctx.connections):
+            c.connect()
+        await settle(loop)
+
+        # hold 6 claims for a while so the LPF sees high busy load
+        held = []
+
+        def keep(err, hdl=None, conn=None):
+            if err is None:
+                held.append(hdl)
+
+        for _ in range(6):
+            ctx.pool.claim({}, keep)
+        for _ in range(75):
+            # step under the 500ms connect timeout so fresh slots get
+            # connected before they expire
+            await advance(loop, 0.2)
+            for c in list(ctx.connections):
+                if not c.connected and not c.dead:
+                    c.connect()
+        assert len(held) == 6
+        high = ctx.pool.get_stats()["totalConnections"]
+        assert high >= 6
+
+        for h in held:
+            h.release()
+        # immediately after release the pool must NOT have dropped to
+        # spares=1: the low-pass filter clamps the shrink
+        await advance(loop, 1.0)
+        assert ctx.pool.get_stats()["totalConnections"] >= 4
+
+        # after the filter decays (~30s), the pool drains to spares
+        for _ in range(80):
+            await advance(loop, 1.0)
+        assert ctx.pool.get_stats()["totalConnections"] <= 2
+
+        ctx.pool.stop()
+        await advance(loop, 2.0)
+
+    run_vt(lambda loop: body(loop))
