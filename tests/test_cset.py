@@ -345,3 +345,83 @@ def test_set_target_grows_set():
         assert len(ctx.connections) == 3
 
     run_vt(lambda loop: body(loop))
+
+
+def test_removing_last_backends_resolver():
+    """All advertised backends removed at once: everything drains, and
+    the remaining backend takes over (test/cset.test.js:578)."""
+    async def body(loop):
+        ctx = Ctx(loop, target=3, maximum=5)
+        ctx.resolver.start()
+        for b in ("b1", "b2", "b3", "b4"):
+            ctx.resolver.add(b, {})
+        ctx.cset.cs_keys.sort()
+        assert ctx.cset.cs_keys == ["b1", "b2", "b3", "b4"]
+        await settle(loop)
+        assert ctx.counts() == {"b1": 1, "b2": 1, "b3": 1}
+        c1 = ctx.by_backend("b1")[0]
+        c2 = ctx.by_backend("b2")[0]
+        c3 = ctx.by_backend("b3")[0]
+        for c in (c1, c2, c3):
+            c.connect()
+        await advance(loop, 0.4)
+        assert len(ctx.in_set()) == 3
+
+        ctx.resolver.remove("b1")
+        ctx.resolver.remove("b2")
+        ctx.resolver.remove("b3")
+        await advance(loop, 0.4)
+        assert c1.dead and c2.dead and c3.dead
+        assert c1.seen and c2.seen and c3.seen
+        assert len(ctx.in_set()) == 0
+        assert ctx.counts() == {"b4": 1}
+        ctx.by_backend("b4")[0].connect()
+        await advance(loop, 1.0)
+        assert ctx.counts() == {"b4": 1}
+        assert len(ctx.in_set()) == 1
+
+        ctx.cset.stop()
+        await advance(loop, 2.0)
+        assert ctx.cset.is_in_state("stopped")
+
+    run_vt(lambda loop: body(loop))
+
+
+def test_removing_last_backend_rebal():
+    """A rebalance away from an advertised backend drains it only after
+    replacements connect (test/cset.test.js:669)."""
+    async def body(loop):
+        ctx = Ctx(loop, target=2, maximum=5)
+        ctx.resolver.start()
+        for b in ("b1", "b2", "b3", "b4"):
+            ctx.resolver.add(b, {})
+        ctx.cset.cs_keys.sort()
+        assert ctx.cset.cs_keys == ["b1", "b2", "b3", "b4"]
+        await settle(loop)
+        assert ctx.counts() == {"b1": 1, "b2": 1}
+        c1 = ctx.by_backend("b1")[0]
+        c2 = ctx.by_backend("b2")[0]
+        c1.connect()
+        c2.connect()
+        await advance(loop, 0.4)
+        assert len(ctx.in_set()) == 2
+
+        ctx.cset.cs_keys.reverse()  # now prefers b4, b3
+        ctx.cset.rebalance()
+        await advance(loop, 0.45)
+        # b3/b4 slots created; their conns connect -> b1/b2 drain
+        for c in list(ctx.connections):
+            if not c.connected and not c.dead:
+                c.connect()
+        await advance(loop, 1.0)
+        counts = ctx.counts()
+        assert counts.get("b4") == 1 and counts.get("b3") == 1
+        assert "b1" not in counts
+        ins = ctx.in_set()
+        assert len(ins) == 2
+
+        ctx.cset.stop()
+        await advance(loop, 2.0)
+        assert ctx.cset.is_in_state("stopped")
+
+    run_vt(lambda loop: body(loop))
