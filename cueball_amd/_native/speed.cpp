@@ -269,7 +269,18 @@ Emitter_remove_listener(PyObject *self_, PyObject *const *args,
     Py_ssize_t n = PyList_GET_SIZE(ls);
     Py_ssize_t found = -1;
     for (Py_ssize_t i = 0; i < n; i++) {
-        if (PyList_GET_ITEM(ls, i) == args[1]) {
+        PyObject *item = PyList_GET_ITEM(ls, i);
+        if (item == args[1]) {
+            found = i;
+            break;
+        }
+        /* bound methods are fresh objects per attribute access: fall
+         * back to == (matches list.remove semantics in the Python
+         * implementation) */
+        int eq = PyObject_RichCompareBool(item, args[1], Py_EQ);
+        if (eq < 0)
+            return NULL;
+        if (eq) {
             found = i;
             break;
         }
@@ -278,8 +289,17 @@ Emitter_remove_listener(PyObject *self_, PyObject *const *args,
         /* allow removing a once() registration by its inner listener */
         for (Py_ssize_t i = 0; i < n; i++) {
             PyObject *w = PyList_GET_ITEM(ls, i);
-            if (Py_TYPE(w) == &OnceWrapperType &&
-                ((OnceWrapper *)w)->ow_listener == args[1]) {
+            if (Py_TYPE(w) != &OnceWrapperType)
+                continue;
+            PyObject *inner = ((OnceWrapper *)w)->ow_listener;
+            if (inner == args[1]) {
+                found = i;
+                break;
+            }
+            int eq = PyObject_RichCompareBool(inner, args[1], Py_EQ);
+            if (eq < 0)
+                return NULL;
+            if (eq) {
                 found = i;
                 break;
             }

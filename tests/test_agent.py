@@ -270,3 +270,61 @@ def test_pinger_5xx_closes_connection():
         srv.stop()
 
     run(body())
+
+
+def test_no_listener_accumulation_across_keepalive_requests():
+    """Regression: each request's data/close/error listeners must be
+    removed when it completes — bound-method listeners need ==
+    comparison in remove_listener, not identity (native-core bug
+    found by the leak detector under the agent benchmark)."""
+    async def body():
+        srv = MockHttpServer()
+        await srv.start()
+        agent = HttpAgent({
+            "defaultPort": srv.port,
+            "recovery": RECOVERY,
+            "spares": 1,
+            "maximum": 1,
+        })
+        for _ in range(5):
+            resp = await agent.request_async("127.0.0.1", "GET", "/x")
+            assert resp.status_code == 200
+        pool = agent.get_pool("127.0.0.1")
+        socks = [f.get_socket_mgr().sm_socket
+                 for fsms in pool.p_connections.values() for f in fsms]
+        assert socks
+        for sock in socks:
+            # only the socket-manager's own listeners remain
+            assert sock.listener_count("data") == 0
+            assert sock.listener_count("close") <= 1
+            assert sock.listener_count("error") <= 1
+        await stop_agent(agent)
+        srv.stop()
+
+    run(body())
+
+
+def test_remove_listener_bound_method_semantics():
+    from cueball_amd.events import EventEmitter
+
+    class Obj:
+        def __init__(self):
+            self.hits = 0
+
+        def handler(self, *a):
+            self.hits += 1
+
+    em = EventEmitter()
+    o = Obj()
+    em.on("evt", o.handler)   # one bound-method object...
+    em.emit("evt")
+    em.remove_listener("evt", o.handler)  # ...a different, == one
+    em.emit("evt")
+    assert o.hits == 1
+    assert em.listener_count("evt") == 0
+
+    # once() removal by the inner bound method
+    em.once("evt", o.handler)
+    em.remove_listener("evt", o.handler)
+    em.emit("evt")
+    assert o.hits == 1
