@@ -48,8 +48,10 @@ class EventEmitter:
             ls.remove(listener)
         except ValueError:
             # also allow removing a once() registration by its inner fn
+            # (== comparison: bound methods are fresh objects per access)
             for w in ls:
-                if getattr(w, "listener", None) is listener:
+                inner = getattr(w, "listener", None)
+                if inner is not None and inner == listener:
                     ls.remove(w)
                     break
         if not ls:
