@@ -201,6 +201,12 @@ class Agent(EventEmitter):
                                "that already has one.")
         self._add_pool(host, dict(options or {}))
 
+    def create_connection(self, options: Any = None,
+                          connect_listener: Any = None) -> None:
+        """UNIX-socket path: unsupported, as in the reference
+        (lib/agent.js:492-495)."""
+        raise CueballError("UNIX domain sockets not supported")
+
     def is_stopped(self) -> bool:
         return self.cba_stopped
 

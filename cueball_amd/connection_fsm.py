@@ -308,6 +308,17 @@ class ClaimHandle(FSM):
     # events a connection claimer may leak handlers for
     _LEAK_EVENTS = ("close", "error", "readable", "data")
 
+    # hot-path: immutable defaults live on the class; instances only
+    # write them when they change (a claim creates one handle, so every
+    # skipped dict store counts)
+    ch_slot: Optional["ConnectionSlotFSM"] = None
+    ch_release_stack: Optional[List[str]] = None
+    ch_connection: Any = None
+    ch_cancelled = False
+    ch_last_error: Optional[BaseException] = None
+    ch_do_release_leak_check = True
+    ch_pinger = False
+
     def __init__(self, options: Dict[str, Any]) -> None:
         self.ch_claim_timeout = options["claimTimeout"]
         self.ch_pool = options["pool"]
@@ -319,15 +330,7 @@ class ClaimHandle(FSM):
         # hot path: the pool passes a pre-made child logger
         self.ch_log = log if options.get("_logReady") \
             else log.child(component="ClaimHandle")
-
-        self.ch_slot: Optional["ConnectionSlotFSM"] = None
-        self.ch_release_stack: Optional[List[str]] = None
-        self.ch_connection: Any = None
         self.ch_pre_listeners: Dict[str, int] = {}
-        self.ch_cancelled = False
-        self.ch_last_error: Optional[BaseException] = None
-        self.ch_do_release_leak_check = True
-        self.ch_pinger = False
 
         super().__init__("waiting", loop=options.get("loop"))
         self.ch_started = self._loop.time() * 1000.0

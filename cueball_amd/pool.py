@@ -628,6 +628,23 @@ class ConnectionPool(FSM):
             fsm.p_idleq_node = None
             self.rebalance()
 
+    def print_connections(self) -> None:
+        """Debugging helper: dump live connection states and dead set
+        (lib/pool.js:812-832)."""
+        obj: Dict[str, Dict[str, int]] = {}
+        ks = list(self.p_keys)
+        for k in self.p_connections.keys():
+            if k not in ks:
+                ks.append(k)
+        for k in ks:
+            per_state: Dict[str, int] = {}
+            for fsm in self.p_connections.get(k, ()):
+                s = fsm.get_state()
+                per_state[s] = per_state.get(s, 0) + 1
+            obj[k] = per_state
+        print("live:", obj)
+        print("dead:", dict(self.p_dead))
+
     # -- stats / claim ------------------------------------------------------
     def get_stats(self) -> Dict[str, Any]:
         tconns = sum(len(v) for v in self.p_connections.values())
