@@ -43,6 +43,7 @@ PyObject *g_get_loop;          /* python helper: get_loop(loop) */
 PyObject *g_fsm_error;         /* exception class FSMError */
 PyObject *g_entry_name_cache;  /* dict: state name -> "state_x_y" */
 PyObject *g_flush_batches;     /* dict: loop -> _FlushBatch */
+PyObject *g_tracer;            /* optional fn(fsm, state) on transitions */
 
 /* ------------------------------------------------------------------ */
 /* EventEmitter                                                        */
@@ -1098,6 +1099,17 @@ fsm_enter_loop(FSMOb *self, PyObject *state)
         Py_INCREF((PyObject *)scope);
         Py_XSETREF(self->f_scope, (PyObject *)scope);
 
+        if (g_tracer != NULL && g_tracer != Py_None) {
+            PyObject *targs[2] = {(PyObject *)self, target};
+            PyObject *tr = PyObject_Vectorcall(g_tracer, targs, 2, NULL);
+            if (tr == NULL) {
+                Py_DECREF(scope);
+                Py_DECREF(target);
+                return -1;
+            }
+            Py_DECREF(tr);
+        }
+
         PyObject *attr = entry_attr_name(target);
         if (attr == NULL) {
             Py_DECREF(scope);
@@ -1629,7 +1641,21 @@ speed_count_listeners(PyObject *mod, PyObject *const *args,
     return PyLong_FromLong(count);
 }
 
+PyObject *
+speed_set_tracer(PyObject *mod, PyObject *fn)
+{
+    (void)mod;
+    if (fn == Py_None) {
+        Py_CLEAR(g_tracer);
+    } else {
+        Py_INCREF(fn);
+        Py_XSETREF(g_tracer, fn);
+    }
+    Py_RETURN_NONE;
+}
+
 PyMethodDef speed_methods[] = {
+    {"_set_tracer", speed_set_tracer, METH_O, NULL},
     {"count_listeners",
      (PyCFunction)(void (*)(void))speed_count_listeners, METH_FASTCALL,
      NULL},

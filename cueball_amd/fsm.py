@@ -59,6 +59,25 @@ def get_loop(loop: Optional[asyncio.AbstractEventLoop] = None) -> asyncio.Abstra
         return asyncio.get_event_loop()
 
 
+#: optional transition tracer: fn(fsm, new_state) called synchronously on
+#: every state transition — the analog of mooremachine's DTrace probes
+#: on FSM transitions (docs/internals.adoc:125-131).  None => zero cost.
+_TRACER = None
+
+
+def set_transition_tracer(fn) -> None:
+    """Install (or clear, with None) a global FSM transition tracer."""
+    global _TRACER
+    _TRACER = fn
+    if NATIVE:
+        from . import _speed as _sp
+        _sp._set_tracer(fn)
+
+
+def get_transition_tracer():
+    return _TRACER
+
+
 class FSMError(AssertionError):
     """Invalid use of an FSM (bad transition, signal in wrong state...)."""
 
@@ -292,6 +311,8 @@ class FSM(EventEmitter):
                 del hist[0]
             scope = StateScope(self)
             self._fsm_scope = scope
+            if _TRACER is not None:
+                _TRACER(self, target)
             entry = self._entry_for(target)
             self._fsm_entering = True
             try:

@@ -292,3 +292,31 @@ def test_scallback_noop_after_exit():
         assert calls == [1]
 
     run_vt(lambda loop: body(loop))
+
+
+def test_transition_tracer_hook():
+    """set_transition_tracer: the mooremachine-DTrace-probe analog —
+    fires synchronously on every transition of every FSM."""
+    import cueball_amd.fsm as mod_fsm
+
+    seen = []
+
+    def tracer(fsm, state):
+        seen.append((type(fsm).__name__, state))
+
+    async def body(loop):
+        mod_fsm.set_transition_tracer(tracer)
+        try:
+            fsm = Light(loop)
+            fsm.emit("go")
+            fsm.emit("stop")
+        finally:
+            mod_fsm.set_transition_tracer(None)
+        assert [s for (n, s) in seen if n == "Light"] == \
+            ["red", "green", "red"]
+        # cleared: no more events
+        n0 = len(seen)
+        fsm.emit("go")
+        assert len(seen) == n0
+
+    run_vt(lambda loop: body(loop))
