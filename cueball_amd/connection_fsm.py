@@ -318,6 +318,7 @@ class ClaimHandle(FSM):
     ch_last_error: Optional[BaseException] = None
     ch_do_release_leak_check = True
     ch_pinger = False
+    ch_pre_listeners: Any = None
 
     def __init__(self, options: Dict[str, Any]) -> None:
         self.ch_claim_timeout = options["claimTimeout"]
@@ -330,10 +331,26 @@ class ClaimHandle(FSM):
         # hot path: the pool passes a pre-made child logger
         self.ch_log = log if options.get("_logReady") \
             else log.child(component="ClaimHandle")
-        self.ch_pre_listeners: Dict[str, int] = {}
 
         super().__init__("waiting", loop=options.get("loop"))
         self.ch_started = self._loop.time() * 1000.0
+
+    @classmethod
+    def fast(cls, pool: Any, claim_stack: List[str], callback: Callable,
+             log: CueballLogger, claim_timeout: float,
+             loop: Any) -> "ClaimHandle":
+        """Positional constructor for the claim hot path (skips the
+        options-dict plumbing; behavior identical to __init__)."""
+        self = cls.__new__(cls)
+        self.ch_claim_timeout = claim_timeout
+        self.ch_pool = pool
+        self.ch_throw_error = True
+        self.ch_claim_stack = claim_stack
+        self.ch_callback = callback
+        self.ch_log = log
+        FSM.__init__(self, "waiting", loop=loop)
+        self.ch_started = self._loop.time() * 1000.0
+        return self
 
     # -- misuse traps (lib/connection-fsm.js:529-557) -------------------
     @property
@@ -451,6 +468,18 @@ class ClaimHandle(FSM):
         S.valid_transitions(("claimed", "waiting", "cancelled"))
         self.ch_slot.claim(self)
 
+    def _on_claimed_conn_error(self, err: BaseException) -> None:
+        conn = self.ch_connection
+        if count_listeners(conn, "error") == 0 and self.ch_throw_error:
+            # End-user never registered an 'error' listener: surface
+            # the failure loudly (lib/connection-fsm.js:697-706).
+            raise err
+        self.ch_log.warn(
+            "connection emitted error while claimed "
+            '(for claim callback "%s")',
+            getattr(self.ch_callback, "__name__", "?"))
+        self.ch_pool._incr_counter("error-while-claimed")
+
     def state_claimed(self, S: StateScope) -> None:
         S.valid_transitions(("released", "closed"))
 
@@ -459,25 +488,10 @@ class ClaimHandle(FSM):
             return
 
         conn = self.ch_connection
-        self.ch_pre_listeners = {}
-        for evt in self._LEAK_EVENTS:
-            self.ch_pre_listeners[evt] = count_listeners(conn, evt)
-
-        def on_conn_error(err: BaseException) -> None:
-            count = count_listeners(conn, "error")
-            if count == 0 and self.ch_throw_error:
-                # End-user never registered an 'error' listener: surface
-                # the failure loudly (lib/connection-fsm.js:697-706).
-                raise err
-            self.ch_log.warn(
-                "connection emitted error while claimed "
-                '(for claim callback "%s")',
-                getattr(self.ch_callback, "__name__", "?"))
-            self.ch_pool._incr_counter("error-while-claimed")
-
-        on_conn_error._cueball_internal = True  # type: ignore[attr-defined]
-        S.on(conn, "error", on_conn_error)
-
+        cnt = count_listeners
+        self.ch_pre_listeners = (cnt(conn, "close"), cnt(conn, "error"),
+                                 cnt(conn, "readable"), cnt(conn, "data"))
+        S.on(conn, "error", self._on_claimed_conn_error)
         self.ch_callback(None, self, conn)
 
     def state_released(self, S: StateScope) -> None:
@@ -485,13 +499,16 @@ class ClaimHandle(FSM):
         if not self.ch_do_release_leak_check:
             return
         conn = self.ch_connection
-        for evt in self._LEAK_EVENTS:
-            new_count = count_listeners(conn, evt)
-            old_count = self.ch_pre_listeners.get(evt)
-            if old_count is not None and new_count > old_count:
+        pre = self.ch_pre_listeners
+        if pre is None:
+            return
+        cnt = count_listeners
+        for i, evt in enumerate(self._LEAK_EVENTS):
+            new_count = cnt(conn, evt)
+            if new_count > pre[i]:
                 self.ch_log.warn(
                     "connection claimer looks like it leaked event "
-                    "handlers", event=evt, count_before_claim=old_count,
+                    "handlers", event=evt, count_before_claim=pre[i],
                     count_after_release=new_count)
 
     def state_closed(self, S: StateScope) -> None:
@@ -505,6 +522,10 @@ class ClaimHandle(FSM):
     def state_failed(self, S: StateScope) -> None:
         S.valid_transitions([])
         S.immediate(lambda: self.ch_callback(self.ch_last_error))
+
+
+# the handle's own claimed-state error listener is not a user leak
+ClaimHandle._on_claimed_conn_error._cueball_internal = True  # type: ignore[attr-defined]
 
 
 def count_listeners(emitter: Any, event: str) -> int:

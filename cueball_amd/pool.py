@@ -690,16 +690,8 @@ class ConnectionPool(FSM):
                 cb, mod_errors.PoolFailedError(self, self.p_last_error))
 
         stack = mod_utils.maybe_capture_stack_trace()
-
-        handle = ClaimHandle({
-            "pool": self,
-            "claimStack": stack,
-            "callback": cb,
-            "log": self.p_claim_log,
-            "_logReady": True,
-            "claimTimeout": timeout,
-            "loop": self._loop,
-        })
+        handle = ClaimHandle.fast(self, stack, cb, self.p_claim_log,
+                                  timeout, self._loop)
 
         def try_next() -> None:
             if not handle.is_in_state("waiting"):
