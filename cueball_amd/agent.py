@@ -278,71 +278,7 @@ class Agent(EventEmitter):
         if host not in self.pools:
             self._add_pool(host, options)
         pool = self.pools[host]
-
-        state: Dict[str, Any] = {"waiter": None, "conn": None, "sock": None}
-
-        def on_abort() -> None:
-            if state["waiter"] is not None:
-                state["waiter"].cancel()
-                state["waiter"] = None
-            if state["conn"] is not None:
-                sock = state["sock"]
-                sock.remove_listener("close", on_close)
-                sock.remove_listener("free", on_free)
-                sock.remove_listener("agentRemove", on_agent_remove)
-                state["conn"].close()
-                state["conn"] = None
-                state["sock"] = None
-
-        def on_close() -> None:
-            sock = state["sock"]
-            sock.remove_listener("free", on_free)
-            sock.remove_listener("agentRemove", on_agent_remove)
-            req.remove_listener("abort", on_abort)
-            conn = state["conn"]
-            # A 'close' straight after a normally-completed request is
-            # indistinguishable from a premature close; give it the
-            # benefit of the doubt: disable the leak check and release
-            # rather than close (lib/agent.js:328-360).
-            conn.disable_release_leak_check()
-            conn.release()
-            state["conn"] = None
-            state["sock"] = None
-
-        def on_free() -> None:
-            sock = state["sock"]
-            sock.remove_listener("close", on_close)
-            sock.remove_listener("agentRemove", on_agent_remove)
-            req.remove_listener("abort", on_abort)
-            state["conn"].release()
-            state["conn"] = None
-            state["sock"] = None
-
-        def on_agent_remove() -> None:
-            # Upgrade etc.: the socket now belongs to someone else; hold
-            # the lease until 'close'.
-            state["sock"].remove_listener("free", on_free)
-            req.remove_listener("abort", on_abort)
-
-        req.once("abort", on_abort)
-
-        def claimed(err: Optional[BaseException], connh: Any = None,
-                    socket: Any = None) -> None:
-            state["waiter"] = None
-            if err is not None:
-                fakesock = FakeSocket()
-                req.on_socket(fakesock)
-                self._loop.call_soon(lambda: fakesock.emit("error", err))
-                return
-            state["conn"] = connh
-            state["sock"] = socket
-            socket.once("free", on_free)
-            socket.once("close", on_close)
-            socket.once("agentRemove", on_agent_remove)
-            req.on_socket(socket)
-
-        state["waiter"] = pool.claim(
-            {"errorOnEmpty": self.cba_err_on_empty}, claimed)
+        _RequestTicket(self, pool, req)
 
     def request(self, options_or_host: Any, method: str = "GET",
                 path: str = "/", headers: Optional[Dict[str, str]] = None,
@@ -420,6 +356,81 @@ class Agent(EventEmitter):
                      latency=round((self._loop.time() - t1) * 1000, 2)),
             handle.close()))
         agent.add_request(req, {})
+
+
+class _RequestTicket:
+    """One request's claim + socket-event protocol (the closure-free
+    form of lib/agent.js:296-396; this sits on the HTTP hot path)."""
+
+    __slots__ = ("agent", "req", "waiter", "conn", "sock")
+
+    def __init__(self, agent: "Agent", pool: Any, req: HttpRequest) -> None:
+        self.agent = agent
+        self.req = req
+        self.conn: Any = None
+        self.sock: Any = None
+        req.once("abort", self.on_abort)
+        self.waiter = pool.claim(
+            {"errorOnEmpty": agent.cba_err_on_empty}, self.claimed)
+
+    def claimed(self, err: Optional[BaseException], connh: Any = None,
+                socket: Any = None) -> None:
+        self.waiter = None
+        if err is not None:
+            fakesock = FakeSocket()
+            self.req.on_socket(fakesock)
+            self.agent._loop.call_soon(
+                lambda: fakesock.emit("error", err))
+            return
+        self.conn = connh
+        self.sock = socket
+        socket.once("free", self.on_free)
+        socket.once("close", self.on_close)
+        socket.once("agentRemove", self.on_agent_remove)
+        self.req.on_socket(socket)
+
+    def on_abort(self) -> None:
+        if self.waiter is not None:
+            self.waiter.cancel()
+            self.waiter = None
+        if self.conn is not None:
+            sock = self.sock
+            sock.remove_listener("close", self.on_close)
+            sock.remove_listener("free", self.on_free)
+            sock.remove_listener("agentRemove", self.on_agent_remove)
+            self.conn.close()
+            self.conn = None
+            self.sock = None
+
+    def on_close(self) -> None:
+        sock = self.sock
+        sock.remove_listener("free", self.on_free)
+        sock.remove_listener("agentRemove", self.on_agent_remove)
+        self.req.remove_listener("abort", self.on_abort)
+        # A 'close' straight after a normally-completed request is
+        # indistinguishable from a premature close; give it the benefit
+        # of the doubt: disable the leak check and release rather than
+        # close (lib/agent.js:328-360).
+        conn = self.conn
+        conn.disable_release_leak_check()
+        conn.release()
+        self.conn = None
+        self.sock = None
+
+    def on_free(self) -> None:
+        sock = self.sock
+        sock.remove_listener("close", self.on_close)
+        sock.remove_listener("agentRemove", self.on_agent_remove)
+        self.req.remove_listener("abort", self.on_abort)
+        self.conn.release()
+        self.conn = None
+        self.sock = None
+
+    def on_agent_remove(self) -> None:
+        # Upgrade etc.: the socket now belongs to someone else; hold
+        # the lease until 'close'.
+        self.sock.remove_listener("free", self.on_free)
+        self.req.remove_listener("abort", self.on_abort)
 
 
 class PingAgent(EventEmitter):
