@@ -319,10 +319,13 @@ class MockDnsServer:
     def __init__(self) -> None:
         self.zone: Dict[Tuple[str, str], List[Dict[str, Any]]] = {}
         self.queries: List[Tuple[str, str]] = []
+        self.tcp_queries: List[Tuple[str, str]] = []
         self.port: Optional[int] = None
         self.drop_next = 0           # drop this many queries (timeouts)
         self.rcode_override: Optional[str] = None
+        self.truncate_udp = False    # answer UDP with TC=1 (force TCP)
         self._transport = None
+        self._tcp_server: Optional[asyncio.AbstractServer] = None
 
     # -- zone building --------------------------------------------------
     def add_srv(self, name: str, target: str, port: int,
@@ -359,19 +362,40 @@ class MockDnsServer:
         self._transport, _ = await loop.create_datagram_endpoint(
             Proto, local_addr=("127.0.0.1", port))
         self.port = self._transport.get_extra_info("sockname")[1]
+        # TCP listener on the same port (DNS truncation fallback)
+        self._tcp_server = await asyncio.start_server(
+            self._handle_tcp, "127.0.0.1", self.port)
         return self.port
+
+    async def _handle_tcp(self, reader: asyncio.StreamReader,
+                          writer: asyncio.StreamWriter) -> None:
+        try:
+            hdr = await reader.readexactly(2)
+            ln = int.from_bytes(hdr, "big")
+            data = await reader.readexactly(ln)
+            resp = self._handle(data, via_tcp=True)
+            if resp is not None:
+                writer.write(len(resp).to_bytes(2, "big") + resp)
+                await writer.drain()
+        except (asyncio.IncompleteReadError, ConnectionResetError):
+            pass
+        finally:
+            writer.close()
 
     def stop(self) -> None:
         if self._transport is not None:
             self._transport.close()
             self._transport = None
+        if self._tcp_server is not None:
+            self._tcp_server.close()
+            self._tcp_server = None
 
     @property
     def resolver_address(self) -> str:
         return "127.0.0.1@%d" % self.port
 
     # -- request handling -------------------------------------------------
-    def _handle(self, data: bytes) -> Optional[bytes]:
+    def _handle(self, data: bytes, via_tcp: bool = False) -> Optional[bytes]:
         try:
             q = dns_wire.decode_message(data)
         except ValueError:
@@ -383,7 +407,9 @@ class MockDnsServer:
             return None
         question = q.question[0]
         name, rtype = question["name"], question["type"]
-        self.queries.append((name, rtype))
+        (self.tcp_queries if via_tcp else self.queries).append((name, rtype))
+        if self.truncate_udp and not via_tcp:
+            return dns_wire.encode_response(q.id, question, tc=True)
         if self.rcode_override is not None:
             return dns_wire.encode_response(q.id, question,
                                             rcode=self.rcode_override)

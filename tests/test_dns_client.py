@@ -168,3 +168,102 @@ def test_full_resolver_against_real_server():
         srv.stop()
 
     run(body())
+
+
+def test_tcp_fallback_on_truncation():
+    """A TC=1 UDP response makes the client retry the same server over
+    TCP (RFC 1035 §4.2.2)."""
+    async def body():
+        srv = MockDnsServer()
+        await srv.start()
+        srv.truncate_udp = True
+        srv.add_a("big.test", "127.0.0.4")
+        client = DnsClient()
+        msg = await client.lookup_async({
+            "domain": "big.test", "type": "A", "timeout": 3000,
+            "resolvers": [srv.resolver_address]})
+        assert msg.get_answers()[0]["target"] == "127.0.0.4"
+        assert srv.queries, "UDP query must have been tried first"
+        assert srv.tcp_queries == [("big.test", "A")]
+        srv.stop()
+
+    run(body())
+
+
+def test_rcode_voting_on_multierror():
+    """MultiError rcode voting: the resolver adopts the most common
+    rcode across failing nameservers (lib/resolver.js:1230-1259)."""
+    import asyncio
+
+    from cueball_amd.resolver import DNSResolverFSM
+
+    async def body():
+        s1 = MockDnsServer()
+        await s1.start()
+        s1.rcode_override = "REFUSED"
+        s2 = MockDnsServer()
+        await s2.start()
+        s2.rcode_override = "NXDOMAIN"
+        s3 = MockDnsServer()
+        await s3.start()
+        s3.rcode_override = "NXDOMAIN"
+
+        res_fsm = DNSResolverFSM({
+            "domain": "vote.test",
+            "resolvers": [s1.resolver_address, s2.resolver_address,
+                          s3.resolver_address],
+            "recovery": {"default": {"timeout": 2000, "retries": 1,
+                                     "delay": 10}},
+        })
+        req = res_fsm.resolve("vote.test", "A", 2000)
+        errs = []
+        done = asyncio.get_running_loop().create_future()
+        req.on("error", lambda e: (errs.append(e),
+                                   done.done() or done.set_result(None)))
+        req.on("answers", lambda *a: done.done() or done.set_result(None))
+        req.send()
+        await asyncio.wait_for(done, 10)
+        assert len(errs) == 1
+        # NXDOMAIN won the vote 2:1 => converted to NoNameError
+        from cueball_amd.dns_client import NoNameError
+        assert isinstance(errs[0], NoNameError)
+        s1.stop(); s2.stop(); s3.stop()
+
+    run(body())
+
+
+def test_client_concurrency_cap():
+    """The shared client caps in-flight lookups (mname-client's
+    concurrency option, lib/resolver.js:385-392)."""
+    import asyncio
+
+    async def body():
+        srv = MockDnsServer()
+        await srv.start()
+        for i in range(8):
+            srv.add_a("c%d.test" % i, "127.0.0.%d" % (i + 1))
+        client = DnsClient(concurrency=2)
+
+        in_flight = {"now": 0, "max": 0}
+        orig = client._query_one
+
+        async def counted(*args, **kw):
+            in_flight["now"] += 1
+            in_flight["max"] = max(in_flight["max"], in_flight["now"])
+            try:
+                await asyncio.sleep(0.02)
+                return await orig(*args, **kw)
+            finally:
+                in_flight["now"] -= 1
+
+        client._query_one = counted
+        results = await asyncio.gather(*[
+            client.lookup_async({"domain": "c%d.test" % i, "type": "A",
+                                 "timeout": 3000,
+                                 "resolvers": [srv.resolver_address]})
+            for i in range(8)])
+        assert len(results) == 8
+        assert in_flight["max"] <= 2
+        srv.stop()
+
+    run(body())
