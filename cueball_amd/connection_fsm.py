@@ -695,20 +695,40 @@ class ConnectionSlotFSM(FSM):
         # (lib/connection-fsm.js:1053-1057).
         if self.csf_monitor is True:
             self.csf_monitor = False
-            smgr.set_monitor(False)
+            if smgr.is_in_state("connected") or smgr.is_in_state("init"):
+                smgr.set_monitor(False)
+            else:
+                # Divergence from the reference (bug fix, property
+                # suite): the monitor's socket connected and died in
+                # the same loop spin, so the smgr already sits in
+                # error/closed — the reference's setMonitor assert
+                # (lib/connection-fsm.js:176) would crash here.
+                # Convert to a normal slot directly: same outcome as
+                # if the events had arrived one spin apart.
+                smgr.sm_monitor = False
+                smgr.reset_backoff()
 
         def on_unwanted() -> None:
             if smgr.is_in_state("connected"):
                 S.goto_state("stopping")
-
-        if not self.csf_wanted:
-            on_unwanted()
-            return
-        S.on(self, "unwanted", on_unwanted)
+            elif smgr.is_in_state("error") or smgr.is_in_state("closed"):
+                # Divergence from the reference (bug fix, found by the
+                # property suite): a slot flagged unwanted whose socket
+                # connected and then died within the same loop spin
+                # reaches idle with the smgr already in error/closed.
+                # The reference's early-return would leave the slot
+                # wedged in idle with no listeners registered
+                # (lib/connection-fsm.js:1059-1062 registers nothing
+                # when !csf_wanted).  An unwanted slot with a dead
+                # socket is simply done.
+                S.goto_state("stopped")
 
         def on_smgr_state(st: str) -> None:
             if st == "error":
-                S.goto_state("retrying")
+                if self.csf_wanted:
+                    S.goto_state("retrying")
+                else:
+                    S.goto_state("stopped")
             elif st == "closed":
                 if not self.csf_wanted:
                     S.goto_state("stopped")
@@ -718,7 +738,14 @@ class ConnectionSlotFSM(FSM):
                 raise FSMError(
                     'Unhandled smgr state transition: connected => "%s"' % st)
 
+        # register the smgr listener in every case — pending smgr events
+        # must find a handler even when the slot is already unwanted
         S.on(smgr, "stateChanged", on_smgr_state)
+
+        if not self.csf_wanted:
+            on_unwanted()
+            return
+        S.on(self, "unwanted", on_unwanted)
 
         if self.csf_check_timeout is not None and \
                 self.csf_checker is not None:

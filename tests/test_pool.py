@@ -824,3 +824,35 @@ def test_lpf_anti_shrink():
         await advance(loop, 2.0)
 
     run_vt(lambda loop: body(loop))
+
+
+def test_unwanted_slot_socket_dies_same_spin():
+    """Regression (found by the property suite): a slot made unwanted
+    while connecting, whose socket connects and then errors in the same
+    loop spin, must end up stopped — not wedged in idle with the smgr
+    in 'error' and no listeners (a latent gap in the reference's
+    state_idle early return, lib/connection-fsm.js:1059-1062)."""
+    async def body(loop):
+        ctx = Ctx(loop, spares=2, maximum=4,
+                  recovery={"default": {"timeout": 500, "retries": 2,
+                                        "delay": 0}})
+        ctx.add_backend("b0")
+        await advance(loop, 0.05)
+        ctx.add_backend("b1")  # rebalance flags one b0 slot unwanted
+        live = [c for c in ctx.connections
+                if not c.connected and not c.dead]
+        conn = live[0]
+        conn.connect()
+        handled = []
+        conn.once("error", handled.append)
+        conn.emit("error", RuntimeError("prop"))
+        await advance(loop, 2.0)
+
+        ctx.pool.stop()
+        await advance(loop, 6.0)
+        assert ctx.pool.is_in_state("stopped"), {
+            k: [(f.get_state(), f.csf_smgr.get_state())
+                for f in fl]
+            for k, fl in ctx.pool.p_connections.items()}
+
+    run_vt(lambda loop: body(loop))
