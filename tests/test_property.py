@@ -189,3 +189,142 @@ async def _scenario(loop, actions):
     assert pool.is_in_state("stopped")
     stats = pool.get_stats()
     assert stats["totalConnections"] == 0
+
+
+# ---------------------------------------------------------------------------
+# ConnectionSet fuzzing: the added/removed contract under churn
+
+CSET_ACTIONS = st.lists(
+    st.tuples(
+        st.sampled_from([
+            "add_backend", "remove_backend", "connect", "conn_error",
+            "conn_close", "set_target", "close_handle", "tick",
+            "big_tick",
+        ]),
+        st.integers(min_value=0, max_value=7),
+    ),
+    min_size=5,
+    max_size=50,
+)
+
+
+@settings(max_examples=150, deadline=None,
+          suppress_health_check=[HealthCheck.too_slow])
+@given(actions=CSET_ACTIONS)
+def test_cset_survives_random_interleavings(actions):
+    loop = VirtualLoop()
+    try:
+        loop.run_until_complete(_cset_scenario(loop, actions))
+    finally:
+        loop.close()
+
+
+async def _cset_scenario(loop, actions):
+    from cueball_amd.connection_set import ConnectionSet
+
+    conns = []
+    resolver = DummyResolver()
+
+    def ctor(backend):
+        c = DummyConnection(backend)
+        c.key = backend.get("key")
+        conns.append(c)
+        orig = c.destroy
+
+        def destroy():
+            if c in conns:
+                conns.remove(c)
+            orig()
+
+        c.destroy = destroy
+        return c
+
+    cset = ConnectionSet({
+        "constructor": ctor,
+        "recovery": RECOVERY,
+        "target": 2,
+        "maximum": 4,
+        "resolver": resolver,
+        "loop": loop,
+    })
+
+    advertised = {}   # ckey -> (conn, hdl)
+    all_added = []
+    all_removed = []
+
+    def on_added(ckey, conn, hdl):
+        assert ckey not in advertised, "double-advertise of %s" % ckey
+        advertised[ckey] = (conn, hdl)
+        all_added.append(ckey)
+
+    def on_removed(ckey, conn, hdl):
+        assert ckey in advertised, "removed before added: %s" % ckey
+        advertised.pop(ckey)
+        all_removed.append(ckey)
+        hdl.release()
+
+    cset.on("added", on_added)
+    cset.on("removed", on_removed)
+    resolver.start()
+
+    backends = set()
+    next_backend = [0]
+
+    for kind, seed in actions:
+        if kind == "add_backend":
+            k = "c%d" % next_backend[0]
+            next_backend[0] += 1
+            backends.add(k)
+            resolver.add(k, {})
+        elif kind == "remove_backend":
+            if backends:
+                k = sorted(backends)[seed % len(backends)]
+                backends.discard(k)
+                resolver.remove(k)
+        elif kind == "connect":
+            live = [c for c in conns if not c.connected and not c.dead]
+            if live:
+                live[seed % len(live)].connect()
+        elif kind == "conn_error":
+            live = [c for c in conns if not c.dead]
+            if live:
+                c = live[seed % len(live)]
+                c.once("error", lambda e: None)
+                c.emit("error", RuntimeError("prop"))
+        elif kind == "conn_close":
+            live = [c for c in conns if not c.dead]
+            if live:
+                live[seed % len(live)].emit("close")
+        elif kind == "set_target":
+            cset.set_target(1 + seed % 4)
+        elif kind == "close_handle":
+            if advertised:
+                ck = sorted(advertised)[seed % len(advertised)]
+                conn, hdl = advertised.pop(ck)
+                # user may .close() an advertised handle at any time;
+                # the set then stops that logical connection directly —
+                # no 'removed' is emitted for it (lib/set.js:760-767)
+                hdl.close()
+        elif kind == "tick":
+            await advance(loop, 0.05)
+        elif kind == "big_tick":
+            await advance(loop, 0.7)
+
+        # invariants
+        assert len(all_added) == len(set(all_added)), "ckey reuse"
+        # singleton: at most one advertised connection per backend
+        per_backend = {}
+        for ck in advertised:
+            bk = ck.rsplit(".", 1)[0]
+            per_backend[bk] = per_backend.get(bk, 0) + 1
+        assert all(v == 1 for v in per_backend.values()), per_backend
+        assert len(cset.cs_fsm) <= 4 + 1
+
+    cset.stop()
+    await advance(loop, 8.0)
+    assert cset.is_in_state("stopped"), {
+        k: (f.get_state(), f.csf_smgr.get_state(), f.csf_wanted,
+            f.csf_monitor)
+        for k, f in cset.cs_fsm.items()}
+    # everything advertised was eventually removed
+    assert not advertised
