@@ -424,7 +424,7 @@ async def scenario_agent(args, results):
             time.monotonic() < t_deadline:
         await asyncio.sleep(0.01)
 
-    concurrent = 1000
+    concurrent = args.agent_concurrency
     lat = []
 
     async def one_get(i):
@@ -531,6 +531,9 @@ def main():
     ap.add_argument("--spares", type=int, default=8)
     ap.add_argument("--maximum", type=int, default=16)
     ap.add_argument("--churn-interval", type=float, default=5.0)
+    ap.add_argument("--agent-concurrency", type=int, default=1000,
+                    help="concurrent GETs in the agent scenario "
+                         "(BASELINE config #3 says 1000)")
     args = ap.parse_args()
 
     rank, world = dist_init()

@@ -344,3 +344,56 @@ def test_timeout_retries_then_fails():
         await settle(loop)
 
     run_vt(lambda loop: body(loop))
+
+
+def test_resolver_stop_restart():
+    """stop() then start() runs the pipeline again (stop drains to init
+    via the sleep state; lib/resolver.js:1110-1118, :452-463)."""
+    async def body(loop):
+        res, nsc = make_resolver(loop, "a.ok", service="_svc._tcp",
+                                 defaultPort=80)
+        st = collect(res)
+        res.start()
+        await advance(loop, 1.0)
+        assert res.is_in_state("running")
+        assert len(st["added"]) == 1
+        n_queries = len(nsc.history)
+
+        res.stop()
+        await advance(loop, 1.0)
+        assert res.is_in_state("stopped")
+
+        res.start()
+        await advance(loop, 1.0)
+        assert res.is_in_state("running")
+        # pipeline re-ran (fresh SRV query at least)
+        assert len(nsc.history) > n_queries
+        res.stop()
+        await settle(loop)
+
+    run_vt(lambda loop: body(loop))
+
+
+def test_static_resolver_stop_restart():
+    from cueball_amd.resolver import StaticIpResolver
+
+    async def body(loop):
+        res = StaticIpResolver({
+            "backends": [{"address": "10.0.0.1", "port": 80}],
+            "loop": loop,
+        })
+        added = []
+        res.on("added", lambda k, b: added.append(k))
+        res.start()
+        await settle(loop)
+        assert res.is_in_state("running")
+        assert len(added) == 1
+        res.stop()
+        await settle(loop)
+        assert res.is_in_state("stopped")
+        res.start()
+        await settle(loop)
+        assert res.is_in_state("running")
+        assert len(added) == 2  # re-emitted on restart
+
+    run_vt(lambda loop: body(loop))
