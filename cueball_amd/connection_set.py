@@ -559,15 +559,27 @@ class LogicalConnection(FSM):
 
     def state_draining(self, S: StateScope) -> None:
         S.valid_transitions(["stopped"])
+        self.lc_set.cs_connections.pop(self.lc_ckey, None)
+
+        hdl = self.lc_hdl
+        if not hdl.is_in_state("claimed"):
+            # Divergence from the reference (bug fix, property suite):
+            # the user already relinquished the handle (e.g. close()
+            # racing the socket's own death event, both pending in the
+            # same spin).  Emitting 'removed' now would hand the
+            # consumer a dead handle whose release() throws
+            # (lib/set.js:792-811 emits unconditionally).  The
+            # connection is already gone: stop directly.
+            S.goto_state("stopped")
+            return
 
         def on_hdl_state(st: str) -> None:
             if st in ("closed", "released", "cancelled"):
                 S.goto_state("stopped")
 
-        S.on(self.lc_hdl, "stateChanged", on_hdl_state)
-        self.lc_set.cs_connections.pop(self.lc_ckey, None)
+        S.on(hdl, "stateChanged", on_hdl_state)
         self.lc_set.assert_emit("removed", self.lc_ckey, self.lc_conn,
-                                self.lc_hdl)
+                                hdl)
 
     def state_stopped(self, S: StateScope) -> None:
         S.valid_transitions([])
