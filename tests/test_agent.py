@@ -328,3 +328,45 @@ def test_remove_listener_bound_method_semantics():
     em.remove_listener("evt", o.handler)
     em.emit("evt")
     assert o.hits == 1
+
+
+def test_agent_dns_hostname_pool():
+    """A DNS-name host makes the agent build a DNS-SRV resolver pool
+    (_http._tcp service) via resolver_for_ip_or_domain."""
+    async def body():
+        from cueball_amd.resolver import DNSResolverFSM
+        from cueball_amd.testing import MockDnsServer
+
+        loop = asyncio.get_running_loop()
+        DNSResolverFSM._nic_cache = {"lo": [
+            {"family": "IPv4", "address": "127.0.0.1"}]}
+        DNSResolverFSM._nic_cache_updated = loop.time() * 1000.0
+
+        srv = MockHttpServer()
+        await srv.start()
+        dns = MockDnsServer()
+        await dns.start()
+        dns.add_srv("_http._tcp.web.test", "w1.web.test", srv.port,
+                    ttl=60)
+        dns.add_a("w1.web.test", "127.0.0.1", ttl=60)
+
+        agent = HttpAgent({
+            "defaultPort": 80,
+            "recovery": RECOVERY,
+            "spares": 1,
+            "maximum": 2,
+            "resolvers": [dns.resolver_address],
+        })
+        resp = await asyncio.wait_for(
+            agent.request_async("web.test", "GET", "/dns-routed"),
+            timeout=20)
+        assert resp.status_code == 200
+        assert json.loads(resp.body)["path"] == "/dns-routed"
+        # the pool resolved through DNS SRV
+        assert ("_http._tcp.web.test", "SRV") in dns.queries
+
+        await stop_agent(agent)
+        dns.stop()
+        srv.stop()
+
+    run(body())
