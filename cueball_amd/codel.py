@@ -13,8 +13,6 @@ Times are in milliseconds on the event loop's (possibly virtual) clock.
 from __future__ import annotations
 
 import math
-from typing import Optional
-
 from .fsm import get_loop
 
 __all__ = ["ControlledDelay", "CODEL_INTERVAL"]

@@ -17,6 +17,7 @@ from __future__ import annotations
 
 import math
 import random
+import time as mod_time
 import uuid as mod_uuid
 from typing import Any, Callable, Dict, List, Optional
 
@@ -76,6 +77,19 @@ class FIRFilter:
             if i < 0:
                 i += n
         return acc
+
+
+class _CancelStub:
+    """claim() return value when the pool is stopping/failed: supports
+    only .cancel() (lib/pool.js:895-897)."""
+
+    __slots__ = ("_state",)
+
+    def __init__(self, state):
+        self._state = state
+
+    def cancel(self) -> None:
+        self._state["done"] = True
 
 
 class _IntervalTimer(EventEmitter):
@@ -527,8 +541,7 @@ class ConnectionPool(FSM):
                 rate_delay + 0.01, self.rebalance)
 
         self.p_in_rebalance = False
-        import time as _time
-        self.p_last_rebalance = _time.time()
+        self.p_last_rebalance = mod_time.time()
 
     # -- slot lifecycle ----------------------------------------------------
     def add_connection(self, key: str) -> None:
@@ -736,12 +749,7 @@ class ConnectionPool(FSM):
             state["done"] = True
 
         self._loop.call_soon(fire)
-
-        class _Stub:
-            def cancel(self) -> None:
-                state["done"] = True
-
-        return _Stub()
+        return _CancelStub(state)
 
     async def claim_async(self, options: Any = None):
         """Coroutine sugar over claim(): returns (handle, connection)."""
