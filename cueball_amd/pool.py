@@ -150,6 +150,15 @@ class ConnectionPool(FSM):
         self.p_claim_log = self.p_log.child(component="ClaimHandle")
 
         self.p_collector = mod_utils.create_error_metrics(options)
+        try:
+            self._gauge = self.p_collector.gauge(
+                name="cueball_pool_connections",
+                help="Live connection counts per pool and state")
+            self._gauge_labels = {"pool": self.p_uuid.split("-")[0],
+                                  "domain": options["domain"]}
+        except (AttributeError, ValueError):
+            self._gauge = None  # foreign collector without gauges
+            self._gauge_labels = {}
 
         spares = options.get("spares")
         maximum = options.get("maximum")
@@ -234,6 +243,15 @@ class ConnectionPool(FSM):
         spares = len(self.p_idleq) + len(self.p_initq)
         busy = conns - spares
         self.p_lpf.put(busy + self.p_spares)
+        # piggyback live gauges on the 5 Hz sample (beyond the
+        # reference, which exposes state through kang only)
+        g = self._gauge
+        if g is not None:
+            labels = self._gauge_labels
+            g.set(conns, {**labels, "state": "total"})
+            g.set(len(self.p_idleq), {**labels, "state": "idle"})
+            g.set(len(self.p_initq), {**labels, "state": "pending"})
+            g.set(len(self.p_waiters), {**labels, "state": "waiting"})
         if self.p_last_rebal_clamped:
             self.rebalance()
 
@@ -351,6 +369,15 @@ class ConnectionPool(FSM):
 
     def state_stopping(self, S: StateScope) -> None:
         S.valid_transitions(["stopping.backends"])
+        # Divergence from the reference (bug fix): fail queued waiters
+        # with PoolStoppingError instead of leaving them pending forever
+        # (the reference's state_stopping never touches p_waiters, so an
+        # infinite-timeout claim outstanding at stop() hangs;
+        # lib/pool.js:433-448 vs the failed-state drain at :398-405).
+        while not self.p_waiters.is_empty():
+            hdl = self.p_waiters.shift()
+            if hdl.is_in_state("waiting"):
+                hdl.fail(mod_errors.PoolStoppingError(self))
         if self.p_started_resolver:
             def on_res_state(s: str) -> None:
                 if s == "stopped":

@@ -370,3 +370,39 @@ def test_agent_dns_hostname_pool():
         srv.stop()
 
     run(body())
+
+
+def test_agent_stop_with_inflight_request():
+    """Stopping the agent fails queued claims with PoolStoppingError;
+    an in-flight request either completes or errors — never hangs."""
+    async def body():
+        srv = MockHttpServer()
+        await srv.start()
+        agent = HttpAgent({
+            "defaultPort": srv.port,
+            "recovery": RECOVERY,
+            "spares": 1,
+            "maximum": 1,
+        })
+        # occupy the single connection
+        resp = await agent.request_async("127.0.0.1", "GET", "/a")
+        assert resp.status_code == 200
+
+        # queue a request, then stop the agent in the same breath
+        fut = asyncio.get_running_loop().create_future()
+
+        def cb(err, resp=None):
+            if not fut.done():
+                fut.set_result((err, resp))
+
+        agent.request("127.0.0.1", "GET", "/b", cb=cb)
+        stop_fut = asyncio.get_running_loop().create_future()
+        agent.stop(lambda e: stop_fut.set_result(None))
+        err, resp2 = await asyncio.wait_for(fut, timeout=15)
+        # either it squeaked through before the stop or it failed
+        # cleanly — both acceptable; hanging is not
+        assert err is not None or resp2.status_code == 200
+        await asyncio.wait_for(stop_fut, timeout=15)
+        srv.stop()
+
+    run(body())

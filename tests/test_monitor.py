@@ -209,3 +209,35 @@ def test_metrics_endpoint():
         kang.stop()
 
     run(body())
+
+
+def test_pool_live_gauges():
+    """Pools publish live connection gauges through the shared metrics
+    collector (piggybacked on the 5 Hz LPF tick; beyond the reference's
+    kang-only state exposure)."""
+    async def body():
+        from cueball_amd.metrics import create_collector
+        from cueball_amd.testing import MockHttpServer
+
+        srv = MockHttpServer()
+        await srv.start()
+        collector = create_collector(labels={"component": "cueball"})
+        agent = HttpAgent({
+            "defaultPort": srv.port,
+            "recovery": RECOVERY,
+            "spares": 2,
+            "maximum": 4,
+            "collector": collector,
+        })
+        await agent.request_async("127.0.0.1", "GET", "/warm")
+        await asyncio.sleep(0.5)  # a couple of LPF ticks
+        text = collector.collect()
+        assert "cueball_pool_connections" in text
+        assert 'state="idle"' in text
+
+        fut = asyncio.get_running_loop().create_future()
+        agent.stop(lambda e: fut.set_result(None))
+        await fut
+        srv.stop()
+
+    run(body())
