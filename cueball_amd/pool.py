@@ -736,6 +736,17 @@ class ConnectionPool(FSM):
         def try_next() -> None:
             if not handle.is_in_state("waiting"):
                 return
+            # The first try runs on the next loop turn; the pool may
+            # have started stopping (or failed) in between — fail now
+            # rather than queueing a waiter nothing will ever feed
+            # (companion to the stopping-state waiter drain).
+            if self.is_in_state("stopping") or self.is_in_state("stopped"):
+                handle.fail(mod_errors.PoolStoppingError(self))
+                return
+            if self.is_in_state("failed"):
+                handle.fail(mod_errors.PoolFailedError(
+                    self, self.p_last_error))
+                return
             # Idle connections sitting around?  Take one.  Entries may be
             # stale ('stateChanged' is async): just unlink and skip them;
             # the slot dispatcher copes (lib/pool.js:934-951).
