@@ -9,6 +9,7 @@ FSM runtime and DNS protocol engine.
 """
 
 from .codel import ControlledDelay
+from .connection import TcpConnection, tcp_constructor
 from .connection_fsm import ClaimHandle, ConnectionSlotFSM
 from .errors import (ClaimHandleMisusedError, ClaimTimeoutError,
                      ConnectionClosedError, ConnectionError_,
@@ -16,6 +17,8 @@ from .errors import (ClaimHandleMisusedError, ClaimTimeoutError,
                      PoolFailedError, PoolStoppingError)
 from .events import EventEmitter
 from .fsm import FSM, FSMError
+from .kang import KangServer
+from .metrics import Collector, create_collector
 from .pool import ConnectionPool
 from .pool_monitor import monitor as pool_monitor
 from .queue import Queue
@@ -58,6 +61,11 @@ __all__ = [
     "ConnectionError_",
     "ConnectionTimeoutError",
     "ConnectionClosedError",
+    "TcpConnection",
+    "tcp_constructor",
+    "KangServer",
+    "Collector",
+    "create_collector",
     "HttpAgent",
     "HttpsAgent",
 ]
