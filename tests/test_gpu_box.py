@@ -62,5 +62,9 @@ def test_bench_short_run():
     line = out.stdout.strip().splitlines()[-1]
     data = json.loads(line)
     assert data["metric"].startswith("pool claims/sec")
-    assert data["value"] > 1000, data
+    # perf floor: the box measures ~100k claims/s with the native core
+    # (profiles/bench_final_headline.json); 30k guards against silent
+    # regressions (e.g. falling back to pure Python) with headroom for
+    # noise
+    assert data["value"] > 30000, data
     assert data["config"]["claim_latency_p50_ms"] is not None
