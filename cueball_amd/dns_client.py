@@ -82,6 +82,9 @@ class NoRecordsError(DnsError):
 
 
 class _UdpProtocol(asyncio.DatagramProtocol):
+    """Delivers datagrams to a replaceable future: the client re-arms
+    ``fut`` after ignoring spoofed/garbage datagrams."""
+
     def __init__(self, fut: asyncio.Future) -> None:
         self.fut = fut
 
@@ -177,10 +180,9 @@ class DnsClient:
     async def _udp_round(self, loop, server: str, port: int, query: bytes,
                          qid: int, deadline: float, domain: str):
         fut: asyncio.Future = loop.create_future()
-        family = 0
+        proto = _UdpProtocol(fut)
         transport, _ = await loop.create_datagram_endpoint(
-            lambda: _UdpProtocol(fut), remote_addr=(server, port),
-            family=family)
+            lambda: proto, remote_addr=(server, port), family=0)
         try:
             transport.sendto(query)
             while True:
@@ -191,9 +193,14 @@ class DnsClient:
                     data = await asyncio.wait_for(fut, timeout=left)
                 except asyncio.TimeoutError:
                     raise TimeoutError_(domain, server) from None
-                msg = dns_wire.decode_message(data)
-                if msg.id != qid:
-                    fut = loop.create_future()  # spoof/stale: keep waiting
+                try:
+                    msg = dns_wire.decode_message(data)
+                except ValueError:
+                    msg = None  # garbage datagram: ignore, keep waiting
+                if msg is None or msg.id != qid:
+                    # spoofed/stale/garbage: re-arm and keep waiting
+                    fut = loop.create_future()
+                    proto.fut = fut
                     continue
                 return msg
         finally:

@@ -158,28 +158,35 @@ def _decode_rr(buf: bytes, off: int) -> Tuple[Dict[str, Any], int]:
 
 
 def decode_message(buf: bytes) -> DnsMessage:
+    """Decode; malformed/truncated input raises ValueError (never a
+    bare struct.error — the client relies on the exception type to
+    treat garbage datagrams as noise)."""
     if len(buf) < 12:
         raise ValueError("DNS message too short")
-    msg = DnsMessage()
-    (msg.id, msg.flags, qd, an, ns, ar) = struct.unpack(">HHHHHH", buf[:12])
-    msg.rcode = msg.flags & 0x0F
-    off = 12
-    for _ in range(qd):
-        qname, off = _decode_name(buf, off)
-        qtype, qclass = struct.unpack(">HH", buf[off:off + 4])
-        off += 4
-        msg.question.append({"name": qname,
-                             "type": TYPE_NAMES.get(qtype, str(qtype)),
-                             "class": qclass})
-    for _ in range(an):
-        rr, off = _decode_rr(buf, off)
-        msg.answers.append(rr)
-    for _ in range(ns):
-        rr, off = _decode_rr(buf, off)
-        msg.authority.append(rr)
-    for _ in range(ar):
-        rr, off = _decode_rr(buf, off)
-        msg.additionals.append(rr)
+    try:
+        msg = DnsMessage()
+        (msg.id, msg.flags, qd, an, ns, ar) = struct.unpack(
+            ">HHHHHH", buf[:12])
+        msg.rcode = msg.flags & 0x0F
+        off = 12
+        for _ in range(qd):
+            qname, off = _decode_name(buf, off)
+            qtype, qclass = struct.unpack(">HH", buf[off:off + 4])
+            off += 4
+            msg.question.append({"name": qname,
+                                 "type": TYPE_NAMES.get(qtype, str(qtype)),
+                                 "class": qclass})
+        for _ in range(an):
+            rr, off = _decode_rr(buf, off)
+            msg.answers.append(rr)
+        for _ in range(ns):
+            rr, off = _decode_rr(buf, off)
+            msg.authority.append(rr)
+        for _ in range(ar):
+            rr, off = _decode_rr(buf, off)
+            msg.additionals.append(rr)
+    except struct.error as e:
+        raise ValueError("truncated DNS message: %s" % e) from e
     return msg
 
 
