@@ -225,7 +225,8 @@ class HttpRequest(EventEmitter):
     def __init__(self, method: str, path: str,
                  headers: Optional[Dict[str, str]] = None,
                  body: Optional[bytes] = None,
-                 host: Optional[str] = None) -> None:
+                 host: Optional[str] = None,
+                 streaming: bool = False) -> None:
         super().__init__()
         self.method = method.upper()
         self.path = path
@@ -237,6 +238,13 @@ class HttpRequest(EventEmitter):
         self._parser: Optional[_ResponseParser] = None
         self._response: Optional[HttpResponse] = None
         self._finished = False
+        # streaming upload (node ClientRequest write()/end() parity):
+        # body chunks written before the socket arrives are buffered
+        self.streaming = streaming
+        self._ended = not streaming
+        self._pending_chunks: List[bytes] = []
+        self._head_sent = False
+        self._chunked_upload = False
 
     # -- user API -------------------------------------------------------
     def abort(self) -> None:
@@ -244,6 +252,49 @@ class HttpRequest(EventEmitter):
             return
         self.aborted = True
         self.emit("abort")
+
+    def write(self, data: bytes) -> None:
+        """Stream a body chunk (requires streaming=True).  Uses chunked
+        transfer-encoding unless Content-Length was given up front."""
+        if not self.streaming:
+            raise RuntimeError("HttpRequest.write() requires streaming=True")
+        if self._ended:
+            raise RuntimeError("write() after end()")
+        if self.conn is None or not self._head_sent:
+            self._pending_chunks.append(data)
+        else:
+            self._send_chunk(data)
+
+    def end(self, data: Optional[bytes] = None) -> None:
+        """Finish a streaming request body."""
+        if not self.streaming or self._ended:
+            if data:
+                raise RuntimeError("end(data) on a non-streaming request")
+            self._ended = True
+            return
+        if data:
+            self.write(data)
+        self._ended = True
+        if self.conn is not None and self._head_sent:
+            self._send_trailer()
+
+    def _send_chunk(self, data: bytes) -> None:
+        if not data:
+            return
+        try:
+            if self._chunked_upload:
+                self.conn.write(b"%x\r\n" % len(data) + data + b"\r\n")
+            else:
+                self.conn.write(data)
+        except (ConnectionResetError, OSError) as e:
+            self._fail(e)
+
+    def _send_trailer(self) -> None:
+        if self._chunked_upload:
+            try:
+                self.conn.write(b"0\r\n\r\n")
+            except (ConnectionResetError, OSError) as e:
+                self._fail(e)
 
     # -- agent protocol ---------------------------------------------------
     def on_socket(self, conn: Any) -> None:
@@ -262,6 +313,13 @@ class HttpRequest(EventEmitter):
 
         try:
             conn.write(self._serialize())
+            self._head_sent = True
+            if self.streaming:
+                for chunk in self._pending_chunks:
+                    self._send_chunk(chunk)
+                self._pending_chunks = []
+                if self._ended:
+                    self._send_trailer()
         except (ConnectionResetError, OSError) as e:
             self._fail(e)
 
@@ -273,7 +331,15 @@ class HttpRequest(EventEmitter):
         if "connection" not in hdrs:
             hdrs["connection"] = ("Connection", "keep-alive")
         body = self.body or b""
-        if body and "content-length" not in hdrs:
+        if self.streaming:
+            if body:
+                raise RuntimeError("streaming request cannot also have "
+                                   "a fixed body")
+            if "content-length" not in hdrs and \
+                    "transfer-encoding" not in hdrs:
+                hdrs["transfer-encoding"] = ("Transfer-Encoding", "chunked")
+                self._chunked_upload = True
+        elif body and "content-length" not in hdrs:
             hdrs["content-length"] = ("Content-Length", str(len(body)))
         for key, (name, value) in hdrs.items():
             lines.append("%s: %s" % (name, value))

@@ -263,16 +263,30 @@ class MockHttpServer:
                         break
                     name, _, value = hline.decode("latin-1").partition(":")
                     headers[name.strip().lower()] = value.strip()
-                clen = int(headers.get("content-length", "0"))
-                if clen:
-                    await reader.readexactly(clen)
+                req_body = b""
+                if "chunked" in headers.get("transfer-encoding", "").lower():
+                    while True:
+                        szline = await reader.readline()
+                        size = int(szline.split(b";")[0].strip() or b"0", 16)
+                        if size == 0:
+                            await reader.readline()  # trailing CRLF
+                            break
+                        req_body += await reader.readexactly(size)
+                        await reader.readexactly(2)  # CRLF
+                else:
+                    clen = int(headers.get("content-length", "0"))
+                    if clen:
+                        req_body = await reader.readexactly(clen)
 
                 self.request_count += 1
                 my_count += 1
                 self.requests_per_conn[slot] = my_count
 
                 close = headers.get("connection", "").lower() == "close"
-                if path == "/ping":
+                if path == "/echo":
+                    body = req_body
+                    status = b"200 OK"
+                elif path == "/ping":
                     self.ping_count += 1
                     body = b"pong"
                     status = b"200 OK"

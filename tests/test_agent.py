@@ -406,3 +406,61 @@ def test_agent_stop_with_inflight_request():
         srv.stop()
 
     run(body())
+
+
+def test_streaming_request_body_chunked():
+    """HttpRequest streaming uploads: write()/end() with chunked
+    transfer-encoding (node ClientRequest parity)."""
+    async def body():
+        from cueball_amd.http_client import HttpRequest
+
+        srv = MockHttpServer()
+        await srv.start()
+        agent = HttpAgent({
+            "defaultPort": srv.port,
+            "recovery": RECOVERY,
+            "spares": 1,
+            "maximum": 1,
+        })
+        req = HttpRequest("POST", "/echo", host="127.0.0.1",
+                          streaming=True)
+        fut = asyncio.get_running_loop().create_future()
+
+        def on_resp(resp):
+            if resp.complete:
+                fut.set_result(resp)
+            else:
+                resp.on("end", lambda: fut.set_result(resp))
+
+        req.on("response", on_resp)
+        req.on("error", lambda e: fut.done() or fut.set_exception(e))
+
+        # write a chunk BEFORE the socket is even claimed (buffered)
+        req.write(b"hello ")
+        agent.add_request(req, "127.0.0.1")
+        await asyncio.sleep(0.05)
+        # and more after the head went out
+        req.write(b"streaming ")
+        req.end(b"world")
+
+        resp = await asyncio.wait_for(fut, timeout=15)
+        assert resp.status_code == 200
+        assert resp.body == b"hello streaming world"
+
+        # the connection survives for keep-alive reuse
+        resp2 = await agent.request_async("127.0.0.1", "GET", "/after")
+        assert resp2.status_code == 200
+        assert srv.conn_count == 1
+
+        await stop_agent(agent)
+        srv.stop()
+
+    run(body())
+
+
+def test_streaming_write_requires_flag():
+    from cueball_amd.http_client import HttpRequest
+
+    req = HttpRequest("POST", "/x", host="h")
+    with pytest.raises(RuntimeError):
+        req.write(b"nope")
