@@ -185,6 +185,7 @@ class DnsClient:
             lambda: proto, remote_addr=(server, port), family=0)
         try:
             transport.sendto(query)
+            resends = 2
             while True:
                 left = deadline - loop.time()
                 if left <= 0:
@@ -193,6 +194,19 @@ class DnsClient:
                     data = await asyncio.wait_for(fut, timeout=left)
                 except asyncio.TimeoutError:
                     raise TimeoutError_(domain, server) from None
+                except OSError:
+                    # connected-UDP sockets surface async ICMP errors
+                    # (port unreachable etc.); these can be stale
+                    # artifacts of a reused ephemeral port — resend a
+                    # bounded number of times before giving up
+                    if resends <= 0:
+                        raise
+                    resends -= 1
+                    await asyncio.sleep(min(0.05, max(0.0, left)))
+                    fut = loop.create_future()
+                    proto.fut = fut
+                    transport.sendto(query)
+                    continue
                 try:
                     msg = dns_wire.decode_message(data)
                 except ValueError:
