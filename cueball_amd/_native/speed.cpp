@@ -44,6 +44,7 @@ PyObject *g_fsm_error;         /* exception class FSMError */
 PyObject *g_entry_name_cache;  /* dict: state name -> "state_x_y" */
 PyObject *g_flush_batches;     /* dict: loop -> _FlushBatch */
 PyObject *g_tracer;            /* optional fn(fsm, state) on transitions */
+PyObject *g_remove_desc;       /* EmitterType's remove_listener descriptor */
 
 /* ------------------------------------------------------------------ */
 /* EventEmitter                                                        */
@@ -725,10 +726,15 @@ Scope_dispose(PyObject *self_, PyObject *noargs)
             PyObject *evt = PyList_GET_ITEM(ls, i + 1);
             PyObject *cb = PyList_GET_ITEM(ls, i + 2);
             PyObject *r;
-            if (Py_TYPE(em) == &EmitterType ||
-                (PyType_IsSubtype(Py_TYPE(em), &EmitterType) &&
-                 /* only safe if remove_listener not overridden; our
-                  * subclasses never override it */ 1)) {
+            /* fast path only when the type's MRO resolves
+             * remove_listener to our own C method (no override) */
+            int fast = 0;
+            if (PyType_IsSubtype(Py_TYPE(em), &EmitterType)) {
+                PyObject *desc = _PyType_Lookup(Py_TYPE(em),
+                                                s_remove_listener);
+                fast = (desc == g_remove_desc);
+            }
+            if (fast) {
                 PyObject *cargs[2] = {evt, cb};
                 r = Emitter_remove_listener(em, cargs, 2);
             } else {
@@ -1705,6 +1711,12 @@ PyInit__speed(void)
         PyType_Ready(&FlushBatchType) < 0 ||
         PyType_Ready(&FSMType) < 0)
         return NULL;
+
+    g_remove_desc = PyDict_GetItemString(EmitterType.tp_dict,
+                                         "remove_listener");
+    if (g_remove_desc == NULL)
+        return NULL;
+    Py_INCREF(g_remove_desc);
 
     PyObject *m = PyModule_Create(&speedmodule);
     if (m == NULL)

@@ -320,3 +320,36 @@ def test_transition_tracer_hook():
         assert len(seen) == n0
 
     run_vt(lambda loop: body(loop))
+
+
+def test_scope_dispose_respects_remove_listener_override():
+    """A user subclass overriding remove_listener must see the removals
+    the state scope performs at exit (the native core's fast path only
+    applies when the MRO resolves to the built-in method)."""
+    removed = []
+
+    class Audited(EventEmitter):
+        def remove_listener(self, event, listener):
+            removed.append(event)
+            super().remove_listener(event, listener)
+
+    async def body(loop):
+        src = Audited()
+
+        class W(FSM):
+            def __init__(self):
+                super().__init__("a", loop=loop)
+
+            def state_a(self, S):
+                S.on(src, "ping", lambda: None)
+                S.on(self, "next", lambda: S.goto_state("b"))
+
+            def state_b(self, S):
+                pass
+
+        w = W()
+        w.emit("next")
+        assert "ping" in removed
+        assert src.listener_count("ping") == 0
+
+    run_vt(lambda loop: body(loop))
