@@ -373,13 +373,24 @@ class MockDnsServer:
                 if resp is not None:
                     self.transport.sendto(resp, addr)
 
-        self._transport, _ = await loop.create_datagram_endpoint(
-            Proto, local_addr=("127.0.0.1", port))
-        self.port = self._transport.get_extra_info("sockname")[1]
-        # TCP listener on the same port (DNS truncation fallback)
-        self._tcp_server = await asyncio.start_server(
-            self._handle_tcp, "127.0.0.1", self.port)
-        return self.port
+        # DNS needs the same port on UDP and TCP; with port=0 the OS
+        # picks a free UDP port whose TCP twin may be taken — retry
+        last_err: Optional[BaseException] = None
+        for _ in range(20):
+            self._transport, _ = await loop.create_datagram_endpoint(
+                Proto, local_addr=("127.0.0.1", port))
+            self.port = self._transport.get_extra_info("sockname")[1]
+            try:
+                self._tcp_server = await asyncio.start_server(
+                    self._handle_tcp, "127.0.0.1", self.port)
+                return self.port
+            except OSError as e:
+                last_err = e
+                self._transport.close()
+                self._transport = None
+                if port != 0:
+                    break
+        raise last_err  # type: ignore[misc]
 
     async def _handle_tcp(self, reader: asyncio.StreamReader,
                           writer: asyncio.StreamWriter) -> None:
