@@ -15,13 +15,22 @@ import pytest
 ROOT = os.path.dirname(os.path.dirname(os.path.abspath(__file__)))
 
 
+def _free_port() -> int:
+    import socket
+    s = socket.socket()
+    s.bind(("127.0.0.1", 0))
+    port = s.getsockname()[1]
+    s.close()
+    return port
+
+
 @pytest.mark.timeout(300)
 def test_bench_two_workers_gloo():
     env = dict(os.environ)
     out = subprocess.run(
         [sys.executable, "-m", "torch.distributed.run", "--nnodes=1",
          "--nproc-per-node", "2", "--master-addr", "127.0.0.1",
-         "--master-port", "29581", "bench.py", "--gpus", "2",
+         "--master-port", str(_free_port()), "bench.py", "--gpus", "2",
          "--steps", "2", "--warmup", "1", "--claims-per-step", "2000"],
         capture_output=True, text=True, timeout=280, cwd=ROOT, env=env)
     assert out.returncode == 0, out.stderr[-3000:]
