@@ -150,3 +150,52 @@ def test_dns_srv_pool_end_to_end():
         await asyncio.sleep(0.1)
 
     run(body())
+
+
+@pytest.mark.timeout(60)
+def test_pool_owns_its_resolver():
+    """No `resolver` option: the pool builds the DNS resolver from
+    domain/resolvers/service, starts it, and stopping the pool stops
+    the resolver too (p_started_resolver path, lib/pool.js:207-221,
+    :433-448)."""
+    async def body():
+        loop = asyncio.get_running_loop()
+        DNSResolverFSM._nic_cache = {"lo": [
+            {"family": "IPv4", "address": "127.0.0.1"}]}
+        DNSResolverFSM._nic_cache_updated = loop.time() * 1000.0
+
+        srv, port = await start_echo()
+        dns = MockDnsServer()
+        await dns.start()
+        dns.add_srv("_own._tcp.own.test", "o1.own.test", port, ttl=30)
+        dns.add_a("o1.own.test", "127.0.0.1", ttl=30)
+
+        pool = ConnectionPool({
+            "domain": "own.test",
+            "service": "_own._tcp",
+            "resolvers": [dns.resolver_address],
+            "constructor": tcp_constructor(loop=loop),
+            "recovery": {"default": {"timeout": 2000, "retries": 3,
+                                     "delay": 100, "maxDelay": 1000}},
+            "spares": 1,
+            "maximum": 2,
+        })
+        # the pool started its own resolver
+        assert pool.p_started_resolver or \
+            pool.p_resolver.is_in_state("stopped")
+
+        assert (await claim_echo(pool)) == port
+        assert pool.p_started_resolver
+        assert pool.p_resolver.is_in_state("running")
+
+        pool.stop()
+        t0 = loop.time()
+        while not pool.is_in_state("stopped") and loop.time() - t0 < 20:
+            await asyncio.sleep(0.05)
+        assert pool.is_in_state("stopped")
+        assert pool.p_resolver.is_in_state("stopped")
+        dns.stop()
+        srv.close()
+        await asyncio.sleep(0.1)
+
+    run(body())
