@@ -18,7 +18,7 @@ import math
 import random
 import time as mod_time
 import uuid as mod_uuid
-from typing import Any, Callable, Dict, List, Optional
+from typing import Any, Dict, List, Optional
 
 from . import utils as mod_utils
 from .connection_fsm import ClaimHandle, ConnectionSlotFSM

@@ -12,7 +12,7 @@ release the claim (mirrors lib/agent.js:322-383).
 
 from __future__ import annotations
 
-from typing import Any, Dict, List, Optional, Tuple
+from typing import Any, Dict, List, Optional
 
 from .events import EventEmitter
 

@@ -25,7 +25,7 @@ import ipaddress
 import math
 import random
 import uuid as mod_uuid
-from typing import Any, Callable, Dict, List, Optional
+from typing import Any, Dict, List, Optional
 
 from . import utils as mod_utils
 from .dns_client import (DnsClient, MultiError, NoNameError, NoRecordsError,
