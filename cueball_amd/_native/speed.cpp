@@ -1560,6 +1560,878 @@ PyTypeObject FSMType = {
     PyType_GenericNew,
 };
 
+
+/* ------------------------------------------------------------------ */
+/* ClaimHandleBase: the pool claim handle's hot path in C              */
+/* ------------------------------------------------------------------ */
+
+PyObject *g_err_claim_timeout;   /* errors.ClaimTimeoutError */
+PyObject *g_err_cueball;         /* errors.CueballError */
+PyObject *g_err_misused;         /* errors.ClaimHandleMisusedError */
+PyObject *g_capture_stack;       /* utils.maybe_capture_stack_trace */
+
+PyObject *s_claim;               /* "claim" */
+PyObject *s_warn;                /* "warn" */
+PyObject *s_incr;                /* "_incr_counter" */
+PyObject *s_claim_timeout_evt;   /* "claim-timeout" */
+PyObject *s_is_in_state;         /* "is_in_state" */
+PyObject *s_time;                /* "time" */
+PyObject *s_waiting, *s_claiming, *s_claimed, *s_released, *s_closed,
+         *s_cancelled, *s_failed, *s_idle, *s_error_evt;
+PyObject *t_waiting_valid;       /* ("claiming","cancelled","failed") */
+PyObject *t_claiming_valid;      /* ("claimed","waiting","cancelled") */
+PyObject *t_claimed_valid;       /* ("released","closed") */
+PyObject *t_empty;               /* () */
+PyObject *s_leak_events[4];      /* close, error, readable, data */
+
+typedef struct {
+    FSMOb base;
+    PyObject *chb_pool;
+    PyObject *chb_claim_stack;
+    PyObject *chb_callback;
+    PyObject *chb_log;
+    PyObject *chb_slot;
+    PyObject *chb_release_stack;
+    PyObject *chb_connection;
+    PyObject *chb_last_error;
+    double chb_claim_timeout;
+    double chb_started;
+    long chb_pre[4];
+    int chb_have_pre;
+    int chb_throw_error;
+    int chb_cancelled;
+    int chb_leak_check;
+    int chb_pinger;
+} CHOb;
+
+extern PyTypeObject CHType;
+
+long
+c_count_listeners(PyObject *emitter, PyObject *event)
+{
+    /* native-emitter path of count_listeners (callers guarantee the
+     * emitter type or fall back in python) */
+    if (!PyObject_TypeCheck(emitter, &EmitterType))
+        return -2;  /* not ours: caller must use python fallback */
+    Emitter *em = (Emitter *)emitter;
+    if (em->ev_events == NULL)
+        return 0;
+    PyObject *ls = PyDict_GetItemWithError(em->ev_events, event);
+    if (ls == NULL)
+        return PyErr_Occurred() ? -1 : 0;
+    long count = 0;
+    Py_ssize_t n = PyList_GET_SIZE(ls);
+    for (Py_ssize_t i = 0; i < n; i++) {
+        PyObject *h = PyList_GET_ITEM(ls, i);
+        if (!PyCallable_Check(h))
+            continue;
+        PyObject *marker = NULL;
+        if (_PyObject_LookupAttr(h, s_internal, &marker) < 0)
+            return -1;
+        if (marker != NULL) {
+            int truthy = PyObject_IsTrue(marker);
+            Py_DECREF(marker);
+            if (truthy < 0)
+                return -1;
+            if (truthy)
+                continue;
+        }
+        PyObject *target = NULL;
+        if (Py_TYPE(h) == &OnceWrapperType) {
+            target = ((OnceWrapper *)h)->ow_listener;
+            Py_INCREF(target);
+        } else if (_PyObject_LookupAttr(h, s_listener, &target) < 0) {
+            return -1;
+        }
+        if (target != NULL && target != h) {
+            if (_PyObject_LookupAttr(target, s_internal, &marker) < 0) {
+                Py_DECREF(target);
+                return -1;
+            }
+            if (marker != NULL) {
+                int truthy = PyObject_IsTrue(marker);
+                Py_DECREF(marker);
+                if (truthy < 0) {
+                    Py_DECREF(target);
+                    return -1;
+                }
+                if (truthy) {
+                    Py_DECREF(target);
+                    continue;
+                }
+            }
+        }
+        Py_XDECREF(target);
+        count++;
+    }
+    return count;
+}
+
+/* claimed-state conn-error listener: raise if the user has no error
+ * listener (and throw_error), else warn + count */
+typedef struct {
+    PyObject_HEAD
+    PyObject *ce_handle;  /* CHOb */
+} ConnErrCb;
+
+extern PyTypeObject ConnErrCbType;
+
+PyObject *
+ConnErrCb_call(PyObject *self_, PyObject *args, PyObject *kwds)
+{
+    CHOb *h = (CHOb *)((ConnErrCb *)self_)->ce_handle;
+    (void)kwds;
+    PyObject *err = (PyTuple_GET_SIZE(args) > 0)
+        ? PyTuple_GET_ITEM(args, 0) : Py_None;
+    long cnt = c_count_listeners(h->chb_connection, s_error_evt);
+    if (cnt == -1)
+        return NULL;
+    if (cnt == 0 && h->chb_throw_error) {
+        /* end-user registered no 'error' listener: surface loudly
+         * (lib/connection-fsm.js:697-706) */
+        if (PyExceptionInstance_Check(err)) {
+            PyErr_SetObject((PyObject *)Py_TYPE(err), err);
+        } else {
+            PyErr_SetObject(PyExc_RuntimeError, err);
+        }
+        return NULL;
+    }
+    PyObject *msg = PyUnicode_FromString(
+        "connection emitted error while claimed");
+    if (msg == NULL)
+        return NULL;
+    PyObject *r = PyObject_CallMethodObjArgs(h->chb_log, s_warn, msg,
+                                             NULL);
+    Py_DECREF(msg);
+    if (r == NULL)
+        return NULL;
+    Py_DECREF(r);
+    PyObject *evt = PyUnicode_FromString("error-while-claimed");
+    if (evt == NULL)
+        return NULL;
+    r = PyObject_CallMethodObjArgs(h->chb_pool, s_incr, evt, NULL);
+    Py_DECREF(evt);
+    if (r == NULL)
+        return NULL;
+    Py_DECREF(r);
+    Py_RETURN_NONE;
+}
+
+int
+ConnErrCb_traverse(PyObject *self_, visitproc visit, void *arg)
+{
+    Py_VISIT(((ConnErrCb *)self_)->ce_handle);
+    return 0;
+}
+
+int
+ConnErrCb_clear_(PyObject *self_)
+{
+    Py_CLEAR(((ConnErrCb *)self_)->ce_handle);
+    return 0;
+}
+
+void
+ConnErrCb_dealloc(PyObject *self_)
+{
+    PyObject_GC_UnTrack(self_);
+    ConnErrCb_clear_(self_);
+    PyObject_GC_Del(self_);
+}
+
+PyObject *
+ConnErrCb_get_internal(PyObject *self_, void *closure)
+{
+    (void)self_; (void)closure;
+    Py_RETURN_TRUE;   /* _cueball_internal marker */
+}
+
+PyGetSetDef ConnErrCb_getset[] = {
+    {(char *)"_cueball_internal", ConnErrCb_get_internal, NULL, NULL, NULL},
+    {NULL, NULL, NULL, NULL, NULL},
+};
+
+PyTypeObject ConnErrCbType = {
+    PyVarObject_HEAD_INIT(NULL, 0)
+    "cueball_amd._speed._ConnErrCb",
+    sizeof(ConnErrCb),
+    0,
+    ConnErrCb_dealloc,
+    0, 0, 0, 0, 0, 0, 0, 0, 0,
+    ConnErrCb_call,
+    0, 0, 0, 0,
+    Py_TPFLAGS_DEFAULT | Py_TPFLAGS_HAVE_GC,
+    0,
+    ConnErrCb_traverse,
+    ConnErrCb_clear_,
+    0, 0, 0, 0, 0, 0,
+    ConnErrCb_getset,
+};
+
+/* failed-state deferred callback: cb(last_error) */
+typedef struct {
+    PyObject_HEAD
+    PyObject *fc_handle;  /* CHOb */
+} FailCb;
+
+extern PyTypeObject FailCbType;
+
+PyObject *
+FailCb_call(PyObject *self_, PyObject *args, PyObject *kwds)
+{
+    CHOb *h = (CHOb *)((FailCb *)self_)->fc_handle;
+    (void)args; (void)kwds;
+    PyObject *err = h->chb_last_error ? h->chb_last_error : Py_None;
+    return PyObject_CallOneArg(h->chb_callback, err);
+}
+
+int
+FailCb_traverse(PyObject *self_, visitproc visit, void *arg)
+{
+    Py_VISIT(((FailCb *)self_)->fc_handle);
+    return 0;
+}
+
+int
+FailCb_clear_(PyObject *self_)
+{
+    Py_CLEAR(((FailCb *)self_)->fc_handle);
+    return 0;
+}
+
+void
+FailCb_dealloc(PyObject *self_)
+{
+    PyObject_GC_UnTrack(self_);
+    FailCb_clear_(self_);
+    PyObject_GC_Del(self_);
+}
+
+PyTypeObject FailCbType = {
+    PyVarObject_HEAD_INIT(NULL, 0)
+    "cueball_amd._speed._FailCb",
+    sizeof(FailCb),
+    0,
+    FailCb_dealloc,
+    0, 0, 0, 0, 0, 0, 0, 0, 0,
+    FailCb_call,
+    0, 0, 0, 0,
+    Py_TPFLAGS_DEFAULT | Py_TPFLAGS_HAVE_GC,
+    0,
+    FailCb_traverse,
+    FailCb_clear_,
+};
+
+/* -- CH helpers ------------------------------------------------------- */
+
+int
+ch_state_is(CHOb *self, PyObject *state)
+{
+    return self->base.f_state == state ||
+        (self->base.f_state != NULL &&
+         PyUnicode_Compare(self->base.f_state, state) == 0);
+}
+
+PyObject *
+ch_err_format(PyObject *cls, const char *msg)
+{
+    PyErr_SetString(cls, msg);
+    return NULL;
+}
+
+/* -- signal methods ---------------------------------------------------- */
+
+PyObject *
+CH_try_(PyObject *self_, PyObject *slot)
+{
+    CHOb *self = (CHOb *)self_;
+    if (!ch_state_is(self, s_waiting))
+        return ch_err_format(g_fsm_error,
+            "ClaimHandle.try_ only in \"waiting\"");
+    PyObject *r = PyObject_CallMethodObjArgs(slot, s_is_in_state, s_idle,
+                                             NULL);
+    if (r == NULL)
+        return NULL;
+    int idle = PyObject_IsTrue(r);
+    Py_DECREF(r);
+    if (!idle)
+        return ch_err_format(g_fsm_error,
+            "ClaimHandle.try_ needs an idle slot");
+    Py_INCREF(slot);
+    Py_XSETREF(self->chb_slot, slot);
+    if (fsm_goto_state(&self->base, s_claiming) < 0)
+        return NULL;
+    Py_RETURN_NONE;
+}
+
+PyObject *
+CH_accept(PyObject *self_, PyObject *connection)
+{
+    CHOb *self = (CHOb *)self_;
+    if (!ch_state_is(self, s_claiming))
+        return ch_err_format(g_fsm_error, "accept only in claiming");
+    Py_INCREF(connection);
+    Py_XSETREF(self->chb_connection, connection);
+    if (fsm_goto_state(&self->base, s_claimed) < 0)
+        return NULL;
+    Py_RETURN_NONE;
+}
+
+PyObject *
+CH_reject(PyObject *self_, PyObject *noargs)
+{
+    CHOb *self = (CHOb *)self_;
+    (void)noargs;
+    if (!ch_state_is(self, s_claiming))
+        return ch_err_format(g_fsm_error, "reject only in claiming");
+    if (fsm_goto_state(&self->base,
+                       self->chb_cancelled ? s_cancelled : s_waiting) < 0)
+        return NULL;
+    Py_RETURN_NONE;
+}
+
+PyObject *CH_release(PyObject *self_, PyObject *noargs);
+
+PyObject *
+CH_cancel(PyObject *self_, PyObject *noargs)
+{
+    CHOb *self = (CHOb *)self_;
+    if (ch_state_is(self, s_claimed))
+        return CH_release(self_, noargs);
+    self->chb_cancelled = 1;
+    /* in "claiming" cancellation applies on reject/accept; in
+     * "waiting" it takes effect now (lib/connection-fsm.js:580) */
+    if (ch_state_is(self, s_waiting)) {
+        if (fsm_goto_state(&self->base, s_cancelled) < 0)
+            return NULL;
+    }
+    Py_RETURN_NONE;
+}
+
+int
+ch_do_claim_timeout(CHOb *self)
+{
+    PyObject *err = PyObject_CallOneArg(g_err_claim_timeout,
+                                        self->chb_pool);
+    if (err == NULL)
+        return -1;
+    Py_XSETREF(self->chb_last_error, err);
+    PyObject *r = PyObject_CallMethodObjArgs(
+        self->chb_pool, s_incr, s_claim_timeout_evt, NULL);
+    if (r == NULL)
+        return -1;
+    Py_DECREF(r);
+    return fsm_goto_state(&self->base, s_failed);
+}
+
+PyObject *
+CH_timeout(PyObject *self_, PyObject *noargs)
+{
+    CHOb *self = (CHOb *)self_;
+    (void)noargs;
+    if (!ch_state_is(self, s_waiting))
+        return ch_err_format(g_fsm_error, "timeout only in waiting");
+    if (ch_do_claim_timeout(self) < 0)
+        return NULL;
+    Py_RETURN_NONE;
+}
+
+PyObject *
+CH__on_claim_timeout(PyObject *self_, PyObject *noargs)
+{
+    return CH_timeout(self_, noargs);
+}
+
+PyObject *
+CH_fail(PyObject *self_, PyObject *err)
+{
+    CHOb *self = (CHOb *)self_;
+    if (ch_state_is(self, s_waiting)) {
+        Py_INCREF(err);
+        Py_XSETREF(self->chb_last_error, err);
+        if (fsm_goto_state(&self->base, s_failed) < 0)
+            return NULL;
+    }
+    Py_RETURN_NONE;
+}
+
+int
+ch_relinquish(CHOb *self, PyObject *target_state)
+{
+    if (!ch_state_is(self, s_claimed)) {
+        if (ch_state_is(self, s_released) || ch_state_is(self, s_closed)) {
+            /* capture ran from C, so the innermost python frame IS
+             * the release() caller (the pure-python twin captures two
+             * frames deeper and indexes -3) */
+            PyObject *by = NULL;
+            PyObject *stack = self->chb_release_stack;
+            if (stack != NULL && PyList_Check(stack) &&
+                PyList_GET_SIZE(stack) >= 1) {
+                by = PyList_GET_ITEM(stack, PyList_GET_SIZE(stack) - 1);
+            }
+            PyObject *msg = PyUnicode_FromFormat(
+                "Connection not claimed by this handle, released by %S",
+                by ? by : Py_None);
+            if (msg == NULL)
+                return -1;
+            PyObject *err = PyObject_CallOneArg(g_err_cueball, msg);
+            Py_DECREF(msg);
+            if (err == NULL)
+                return -1;
+            PyErr_SetObject(g_err_cueball, err);
+            Py_DECREF(err);
+            return -1;
+        }
+        PyObject *msg = PyUnicode_FromFormat(
+            "ClaimHandle.release() called while in state \"%S\"",
+            self->base.f_state ? self->base.f_state : Py_None);
+        if (msg == NULL)
+            return -1;
+        PyErr_SetObject(g_err_cueball, msg);
+        Py_DECREF(msg);
+        return -1;
+    }
+    PyObject *stack = PyObject_CallNoArgs(g_capture_stack);
+    if (stack == NULL)
+        return -1;
+    Py_XSETREF(self->chb_release_stack, stack);
+    return fsm_goto_state(&self->base, target_state);
+}
+
+PyObject *
+CH_release(PyObject *self_, PyObject *noargs)
+{
+    (void)noargs;
+    if (ch_relinquish((CHOb *)self_, s_released) < 0)
+        return NULL;
+    Py_RETURN_NONE;
+}
+
+PyObject *
+CH_close(PyObject *self_, PyObject *noargs)
+{
+    (void)noargs;
+    if (ch_relinquish((CHOb *)self_, s_closed) < 0)
+        return NULL;
+    Py_RETURN_NONE;
+}
+
+PyObject *
+CH_disable_release_leak_check(PyObject *self_, PyObject *noargs)
+{
+    (void)noargs;
+    ((CHOb *)self_)->chb_leak_check = 0;
+    Py_RETURN_NONE;
+}
+
+/* -- state entries ------------------------------------------------------ */
+
+PyObject *
+CH_state_waiting(PyObject *self_, PyObject *scope)
+{
+    CHOb *self = (CHOb *)self_;
+    Py_INCREF(t_waiting_valid);
+    Py_XSETREF(self->base.f_valid, t_waiting_valid);
+    Py_CLEAR(self->chb_slot);
+    if (isfinite(self->chb_claim_timeout)) {
+        PyObject *ms = PyFloat_FromDouble(self->chb_claim_timeout);
+        if (ms == NULL)
+            return NULL;
+        PyObject *cb = PyObject_GetAttrString(self_, "_on_claim_timeout");
+        if (cb == NULL) {
+            Py_DECREF(ms);
+            return NULL;
+        }
+        PyObject *args[2] = {ms, cb};
+        PyObject *r = Scope_timeout(scope, args, 2);
+        Py_DECREF(ms);
+        Py_DECREF(cb);
+        if (r == NULL)
+            return NULL;
+        Py_DECREF(r);
+    }
+    Py_RETURN_NONE;
+}
+
+PyObject *
+CH_state_claiming(PyObject *self_, PyObject *scope)
+{
+    CHOb *self = (CHOb *)self_;
+    (void)scope;
+    Py_INCREF(t_claiming_valid);
+    Py_XSETREF(self->base.f_valid, t_claiming_valid);
+    PyObject *r = PyObject_CallMethodObjArgs(self->chb_slot, s_claim,
+                                             self_, NULL);
+    if (r == NULL)
+        return NULL;
+    Py_DECREF(r);
+    Py_RETURN_NONE;
+}
+
+PyObject *
+CH_state_claimed(PyObject *self_, PyObject *scope)
+{
+    CHOb *self = (CHOb *)self_;
+    Py_INCREF(t_claimed_valid);
+    Py_XSETREF(self->base.f_valid, t_claimed_valid);
+
+    if (self->chb_cancelled) {
+        if (fsm_goto_state(&self->base, s_released) < 0)
+            return NULL;
+        Py_RETURN_NONE;
+    }
+
+    PyObject *conn = self->chb_connection;
+    for (int i = 0; i < 4; i++) {
+        long c = c_count_listeners(conn, s_leak_events[i]);
+        if (c == -1)
+            return NULL;
+        self->chb_pre[i] = (c == -2) ? 0 : c;
+    }
+    self->chb_have_pre = 1;
+
+    ConnErrCb *ec = PyObject_GC_New(ConnErrCb, &ConnErrCbType);
+    if (ec == NULL)
+        return NULL;
+    Py_INCREF(self_);
+    ec->ce_handle = self_;
+    PyObject_GC_Track((PyObject *)ec);
+    PyObject *args[3] = {conn, s_error_evt, (PyObject *)ec};
+    PyObject *r = Scope_on(scope, args, 3);
+    Py_DECREF(ec);
+    if (r == NULL)
+        return NULL;
+    Py_DECREF(r);
+
+    PyObject *cbargs[3] = {Py_None, self_, conn};
+    r = PyObject_Vectorcall(self->chb_callback, cbargs, 3, NULL);
+    if (r == NULL)
+        return NULL;
+    Py_DECREF(r);
+    Py_RETURN_NONE;
+}
+
+PyObject *
+CH_state_released(PyObject *self_, PyObject *scope)
+{
+    CHOb *self = (CHOb *)self_;
+    (void)scope;
+    Py_INCREF(t_empty);
+    Py_XSETREF(self->base.f_valid, t_empty);
+    if (!self->chb_leak_check || !self->chb_have_pre)
+        Py_RETURN_NONE;
+    PyObject *conn = self->chb_connection;
+    for (int i = 0; i < 4; i++) {
+        long c = c_count_listeners(conn, s_leak_events[i]);
+        if (c == -1)
+            return NULL;
+        if (c != -2 && c > self->chb_pre[i]) {
+            PyObject *msg = PyUnicode_FromFormat(
+                "connection claimer looks like it leaked event handlers "
+                "(event=%S before=%ld after=%ld)",
+                s_leak_events[i], self->chb_pre[i], c);
+            if (msg == NULL)
+                return NULL;
+            PyObject *r = PyObject_CallMethodObjArgs(self->chb_log, s_warn,
+                                                     msg, NULL);
+            Py_DECREF(msg);
+            if (r == NULL)
+                return NULL;
+            Py_DECREF(r);
+        }
+    }
+    Py_RETURN_NONE;
+}
+
+PyObject *
+CH_state_closed(PyObject *self_, PyObject *scope)
+{
+    CHOb *self = (CHOb *)self_;
+    (void)scope;
+    Py_INCREF(t_empty);
+    Py_XSETREF(self->base.f_valid, t_empty);
+    Py_RETURN_NONE;
+}
+
+PyObject *
+CH_state_cancelled(PyObject *self_, PyObject *scope)
+{
+    CHOb *self = (CHOb *)self_;
+    (void)scope;
+    Py_INCREF(t_empty);
+    Py_XSETREF(self->base.f_valid, t_empty);
+    Py_RETURN_NONE;
+}
+
+PyObject *
+CH_state_failed(PyObject *self_, PyObject *scope)
+{
+    CHOb *self = (CHOb *)self_;
+    Py_INCREF(t_empty);
+    Py_XSETREF(self->base.f_valid, t_empty);
+    FailCb *fc = PyObject_GC_New(FailCb, &FailCbType);
+    if (fc == NULL)
+        return NULL;
+    Py_INCREF(self_);
+    fc->fc_handle = self_;
+    PyObject_GC_Track((PyObject *)fc);
+    PyObject *r = Scope_immediate(scope, (PyObject *)fc);
+    Py_DECREF(fc);
+    if (r == NULL)
+        return NULL;
+    Py_DECREF(r);
+    Py_RETURN_NONE;
+}
+
+/* -- misuse traps -------------------------------------------------------- */
+
+PyObject *
+CH_get_misused(PyObject *self_, void *closure)
+{
+    (void)self_; (void)closure;
+    PyObject *err = PyObject_CallNoArgs(g_err_misused);
+    if (err == NULL)
+        return NULL;
+    PyErr_SetObject(g_err_misused, err);
+    Py_DECREF(err);
+    return NULL;
+}
+
+PyObject *
+CH_on(PyObject *self_, PyObject *const *args, Py_ssize_t nargs)
+{
+    if (nargs >= 1 && PyUnicode_Check(args[0])) {
+        if (PyUnicode_CompareWithASCIIString(args[0], "readable") == 0 ||
+            PyUnicode_CompareWithASCIIString(args[0], "close") == 0) {
+            PyObject *err = PyObject_CallNoArgs(g_err_misused);
+            if (err == NULL)
+                return NULL;
+            PyErr_SetObject(g_err_misused, err);
+            Py_DECREF(err);
+            return NULL;
+        }
+    }
+    return Emitter_on(self_, args, nargs);
+}
+
+PyObject *
+CH_once(PyObject *self_, PyObject *const *args, Py_ssize_t nargs)
+{
+    if (nargs >= 1 && PyUnicode_Check(args[0])) {
+        if (PyUnicode_CompareWithASCIIString(args[0], "readable") == 0 ||
+            PyUnicode_CompareWithASCIIString(args[0], "close") == 0) {
+            PyObject *err = PyObject_CallNoArgs(g_err_misused);
+            if (err == NULL)
+                return NULL;
+            PyErr_SetObject(g_err_misused, err);
+            Py_DECREF(err);
+            return NULL;
+        }
+    }
+    return Emitter_once(self_, args, nargs);
+}
+
+/* -- init / gc ------------------------------------------------------------ */
+
+int
+CH_traverse(PyObject *self_, visitproc visit, void *arg)
+{
+    CHOb *self = (CHOb *)self_;
+    Py_VISIT(self->chb_pool);
+    Py_VISIT(self->chb_claim_stack);
+    Py_VISIT(self->chb_callback);
+    Py_VISIT(self->chb_log);
+    Py_VISIT(self->chb_slot);
+    Py_VISIT(self->chb_release_stack);
+    Py_VISIT(self->chb_connection);
+    Py_VISIT(self->chb_last_error);
+    return FSM_traverse(self_, visit, arg);
+}
+
+int
+CH_clear_(PyObject *self_)
+{
+    CHOb *self = (CHOb *)self_;
+    Py_CLEAR(self->chb_pool);
+    Py_CLEAR(self->chb_claim_stack);
+    Py_CLEAR(self->chb_callback);
+    Py_CLEAR(self->chb_log);
+    Py_CLEAR(self->chb_slot);
+    Py_CLEAR(self->chb_release_stack);
+    Py_CLEAR(self->chb_connection);
+    Py_CLEAR(self->chb_last_error);
+    return FSM_clear_(self_);
+}
+
+void
+CH_dealloc(PyObject *self_)
+{
+    CHOb *self = (CHOb *)self_;
+    PyTypeObject *tp = Py_TYPE(self_);
+    PyObject_GC_UnTrack(self_);
+    if (self->base.base.ev_weakrefs != NULL)
+        PyObject_ClearWeakRefs(self_);
+    CH_clear_(self_);
+    tp->tp_free(self_);
+}
+
+/* _setup(pool, claim_stack, callback, log, claim_timeout, loop):
+ * one-shot C-side initializer used by the python subclass */
+PyObject *
+CH__setup(PyObject *self_, PyObject *const *args, Py_ssize_t nargs)
+{
+    CHOb *self = (CHOb *)self_;
+    if (nargs != 7) {
+        PyErr_SetString(PyExc_TypeError,
+            "_setup(pool, stack, cb, log, timeout, throw_error, loop)");
+        return NULL;
+    }
+    Py_INCREF(args[0]); Py_XSETREF(self->chb_pool, args[0]);
+    Py_INCREF(args[1]); Py_XSETREF(self->chb_claim_stack, args[1]);
+    Py_INCREF(args[2]); Py_XSETREF(self->chb_callback, args[2]);
+    Py_INCREF(args[3]); Py_XSETREF(self->chb_log, args[3]);
+    self->chb_claim_timeout = PyFloat_AsDouble(args[4]);
+    if (self->chb_claim_timeout == -1.0 && PyErr_Occurred())
+        return NULL;
+    int throw_error = PyObject_IsTrue(args[5]);
+    if (throw_error < 0)
+        return NULL;
+    self->chb_throw_error = throw_error;
+    self->chb_cancelled = 0;
+    self->chb_leak_check = 1;
+    self->chb_pinger = 0;
+    self->chb_have_pre = 0;
+
+    /* FSM init (resolves loop, enters "waiting") */
+    PyObject *resolved = PyObject_CallOneArg(g_get_loop, args[6]);
+    if (resolved == NULL)
+        return NULL;
+    Py_XSETREF(self->base.f_loop, resolved);
+    if (Emitter_init(self_, NULL, NULL) < 0)
+        return NULL;
+    if (self->base.f_emit_queue == NULL) {
+        self->base.f_emit_queue = PyList_New(0);
+        if (self->base.f_emit_queue == NULL)
+            return NULL;
+    }
+    if (self->base.f_history == NULL) {
+        self->base.f_history = PyList_New(0);
+        if (self->base.f_history == NULL)
+            return NULL;
+    }
+    if (fsm_goto_state(&self->base, s_waiting) < 0)
+        return NULL;
+    PyObject *t = PyObject_CallMethodObjArgs(self->base.f_loop, s_time,
+                                             NULL);
+    if (t == NULL)
+        return NULL;
+    double secs = PyFloat_AsDouble(t);
+    Py_DECREF(t);
+    if (secs == -1.0 && PyErr_Occurred())
+        return NULL;
+    self->chb_started = secs * 1000.0;
+    Py_RETURN_NONE;
+}
+
+PyMethodDef CH_methods[] = {
+    {"_setup", (PyCFunction)(void (*)(void))CH__setup, METH_FASTCALL, NULL},
+    {"try_", CH_try_, METH_O, NULL},
+    {"accept", CH_accept, METH_O, NULL},
+    {"reject", CH_reject, METH_NOARGS, NULL},
+    {"cancel", CH_cancel, METH_NOARGS, NULL},
+    {"timeout", CH_timeout, METH_NOARGS, NULL},
+    {"_on_claim_timeout", CH__on_claim_timeout, METH_NOARGS, NULL},
+    {"fail", CH_fail, METH_O, NULL},
+    {"release", CH_release, METH_NOARGS, NULL},
+    {"close", CH_close, METH_NOARGS, NULL},
+    {"disable_release_leak_check", CH_disable_release_leak_check,
+     METH_NOARGS, NULL},
+    {"on", (PyCFunction)(void (*)(void))CH_on, METH_FASTCALL, NULL},
+    {"once", (PyCFunction)(void (*)(void))CH_once, METH_FASTCALL, NULL},
+    {"state_waiting", CH_state_waiting, METH_O, NULL},
+    {"state_claiming", CH_state_claiming, METH_O, NULL},
+    {"state_claimed", CH_state_claimed, METH_O, NULL},
+    {"state_released", CH_state_released, METH_O, NULL},
+    {"state_closed", CH_state_closed, METH_O, NULL},
+    {"state_cancelled", CH_state_cancelled, METH_O, NULL},
+    {"state_failed", CH_state_failed, METH_O, NULL},
+    {NULL, NULL, 0, NULL},
+};
+
+PyMemberDef CH_members[] = {
+    {(char *)"ch_pool", T_OBJECT, offsetof(CHOb, chb_pool), 0, NULL},
+    {(char *)"ch_claim_stack", T_OBJECT, offsetof(CHOb, chb_claim_stack),
+     0, NULL},
+    {(char *)"ch_callback", T_OBJECT, offsetof(CHOb, chb_callback), 0,
+     NULL},
+    {(char *)"ch_log", T_OBJECT, offsetof(CHOb, chb_log), 0, NULL},
+    {(char *)"ch_slot", T_OBJECT, offsetof(CHOb, chb_slot), 0, NULL},
+    {(char *)"ch_release_stack", T_OBJECT,
+     offsetof(CHOb, chb_release_stack), 0, NULL},
+    {(char *)"ch_connection", T_OBJECT, offsetof(CHOb, chb_connection),
+     0, NULL},
+    {(char *)"ch_last_error", T_OBJECT, offsetof(CHOb, chb_last_error),
+     0, NULL},
+    {(char *)"ch_claim_timeout", T_DOUBLE,
+     offsetof(CHOb, chb_claim_timeout), 0, NULL},
+    {(char *)"ch_started", T_DOUBLE, offsetof(CHOb, chb_started), 0, NULL},
+    {(char *)"ch_cancelled", T_INT, offsetof(CHOb, chb_cancelled), 0,
+     NULL},
+    {(char *)"ch_throw_error", T_INT, offsetof(CHOb, chb_throw_error), 0,
+     NULL},
+    {(char *)"ch_do_release_leak_check", T_INT,
+     offsetof(CHOb, chb_leak_check), 0, NULL},
+    {(char *)"ch_pinger", T_INT, offsetof(CHOb, chb_pinger), 0, NULL},
+    {NULL, 0, 0, 0, NULL},
+};
+
+PyGetSetDef CH_getset[] = {
+    {(char *)"readable", CH_get_misused, NULL, NULL, NULL},
+    {(char *)"writable", CH_get_misused, NULL, NULL, NULL},
+    {NULL, NULL, NULL, NULL, NULL},
+};
+
+PyTypeObject CHType = {
+    PyVarObject_HEAD_INIT(NULL, 0)
+    "cueball_amd._speed.ClaimHandleBase",
+    sizeof(CHOb),
+    0,
+    CH_dealloc,
+    0, 0, 0, 0, 0, 0, 0, 0, 0, 0, 0, 0, 0, 0,
+    Py_TPFLAGS_DEFAULT | Py_TPFLAGS_BASETYPE | Py_TPFLAGS_HAVE_GC,
+    "pool claim handle (native core)",
+    CH_traverse,
+    CH_clear_,
+    0, 0, 0, 0,
+    CH_methods,
+    CH_members,
+    CH_getset,
+    &FSMType,
+    0, 0, 0,
+    0,
+    0,
+    0,
+    PyType_GenericNew,
+};
+
+PyObject *
+speed_set_claim_helpers(PyObject *mod, PyObject *const *args,
+                        Py_ssize_t nargs)
+{
+    (void)mod;
+    if (nargs != 4) {
+        PyErr_SetString(PyExc_TypeError,
+            "_set_claim_helpers(ClaimTimeoutError, CueballError, "
+            "ClaimHandleMisusedError, capture_stack)");
+        return NULL;
+    }
+    Py_INCREF(args[0]); Py_XSETREF(g_err_claim_timeout, args[0]);
+    Py_INCREF(args[1]); Py_XSETREF(g_err_cueball, args[1]);
+    Py_INCREF(args[2]); Py_XSETREF(g_err_misused, args[2]);
+    Py_INCREF(args[3]); Py_XSETREF(g_capture_stack, args[3]);
+    Py_RETURN_NONE;
+}
+
 /* ------------------------------------------------------------------ */
 /* module                                                              */
 /* ------------------------------------------------------------------ */
@@ -1661,6 +2533,9 @@ speed_set_tracer(PyObject *mod, PyObject *fn)
 }
 
 PyMethodDef speed_methods[] = {
+    {"_set_claim_helpers",
+     (PyCFunction)(void (*)(void))speed_set_claim_helpers, METH_FASTCALL,
+     NULL},
     {"_set_tracer", speed_set_tracer, METH_O, NULL},
     {"count_listeners",
      (PyCFunction)(void (*)(void))speed_count_listeners, METH_FASTCALL,
@@ -1696,6 +2571,29 @@ PyInit__speed(void)
     s_flush_name = PyUnicode_InternFromString("_flush_state_changed");
     s_internal = PyUnicode_InternFromString("_cueball_internal");
     s_is_closed = PyUnicode_InternFromString("is_closed");
+    s_claim = PyUnicode_InternFromString("claim");
+    s_warn = PyUnicode_InternFromString("warn");
+    s_incr = PyUnicode_InternFromString("_incr_counter");
+    s_claim_timeout_evt = PyUnicode_InternFromString("claim-timeout");
+    s_is_in_state = PyUnicode_InternFromString("is_in_state");
+    s_time = PyUnicode_InternFromString("time");
+    s_waiting = PyUnicode_InternFromString("waiting");
+    s_claiming = PyUnicode_InternFromString("claiming");
+    s_claimed = PyUnicode_InternFromString("claimed");
+    s_released = PyUnicode_InternFromString("released");
+    s_closed = PyUnicode_InternFromString("closed");
+    s_cancelled = PyUnicode_InternFromString("cancelled");
+    s_failed = PyUnicode_InternFromString("failed");
+    s_idle = PyUnicode_InternFromString("idle");
+    s_error_evt = PyUnicode_InternFromString("error");
+    t_waiting_valid = PyTuple_Pack(3, s_claiming, s_cancelled, s_failed);
+    t_claiming_valid = PyTuple_Pack(3, s_claimed, s_waiting, s_cancelled);
+    t_claimed_valid = PyTuple_Pack(2, s_released, s_closed);
+    t_empty = PyTuple_New(0);
+    s_leak_events[0] = PyUnicode_InternFromString("close");
+    s_leak_events[1] = s_error_evt;
+    s_leak_events[2] = PyUnicode_InternFromString("readable");
+    s_leak_events[3] = PyUnicode_InternFromString("data");
     g_entry_name_cache = PyDict_New();
     if (g_entry_name_cache == NULL)
         return NULL;
@@ -1709,6 +2607,9 @@ PyInit__speed(void)
         PyType_Ready(&ScopeType) < 0 ||
         PyType_Ready(&IntervalType) < 0 ||
         PyType_Ready(&FlushBatchType) < 0 ||
+        PyType_Ready(&ConnErrCbType) < 0 ||
+        PyType_Ready(&FailCbType) < 0 ||
+        PyType_Ready(&CHType) < 0 ||
         PyType_Ready(&FSMType) < 0)
         return NULL;
 
@@ -1727,5 +2628,7 @@ PyInit__speed(void)
     PyModule_AddObject(m, "StateScope", (PyObject *)&ScopeType);
     Py_INCREF(&FSMType);
     PyModule_AddObject(m, "FSM", (PyObject *)&FSMType);
+    Py_INCREF(&CHType);
+    PyModule_AddObject(m, "ClaimHandleBase", (PyObject *)&CHType);
     return m;
 }

@@ -528,6 +528,53 @@ class ClaimHandle(FSM):
 ClaimHandle._on_claimed_conn_error._cueball_internal = True  # type: ignore[attr-defined]
 
 
+#: pure-Python claim handle kept importable for the CUEBALL_PURE runtime
+PyClaimHandle = ClaimHandle
+
+if _native_count is not None:
+    # Native core present: the claim handle's entire hot path (signal
+    # methods + the waiting/claiming/claimed/released states) lives in
+    # C (cueball_amd/_native/speed.cpp, ClaimHandleBase).  Semantics
+    # are identical to PyClaimHandle above — the full test suite runs
+    # against both (CUEBALL_PURE=1 selects the Python one).
+    from . import _speed as _speed_mod
+    from ._speed import ClaimHandleBase as _CHBase
+
+    _speed_mod._set_claim_helpers(
+        mod_errors.ClaimTimeoutError,
+        mod_errors.CueballError,
+        mod_errors.ClaimHandleMisusedError,
+        mod_utils.maybe_capture_stack_trace,
+    )
+
+    class ClaimHandle(_CHBase):  # noqa: F811
+        """Claim handle backed by the native core (docs: PyClaimHandle)."""
+
+        def __init__(self, options: Dict[str, Any]) -> None:
+            throw_error = options.get("throwError")
+            log = options["log"]
+            if not options.get("_logReady"):
+                log = log.child(component="ClaimHandle")
+            self._setup(
+                options["pool"],
+                options["claimStack"],
+                options["callback"],
+                log,
+                float(options["claimTimeout"]),
+                True if throw_error is None else bool(throw_error),
+                options.get("loop"),
+            )
+
+        @classmethod
+        def fast(cls, pool: Any, claim_stack: List[str],
+                 callback: Callable, log: CueballLogger,
+                 claim_timeout: float, loop: Any) -> "ClaimHandle":
+            self = cls.__new__(cls)
+            self._setup(pool, claim_stack, callback, log,
+                        float(claim_timeout), True, loop)
+            return self
+
+
 def count_listeners(emitter: Any, event: str) -> int:
     """Count user-registered listeners, ignoring cueball's own internal
     handlers (lib/connection-fsm.js:786-808)."""
