@@ -79,6 +79,61 @@ class FIRFilter:
         return acc
 
 
+class _ClaimTicket:
+    """Per-claim driver: retries try_next whenever the handle returns to
+    'waiting' (registered as the handle's stateChanged listener; the
+    closure-free form of lib/pool.js:922-968, on the claim hot path)."""
+
+    __slots__ = ("pool", "handle", "err_on_empty")
+
+    def __init__(self, pool: "ConnectionPool", handle: ClaimHandle,
+                 err_on_empty: bool) -> None:
+        self.pool = pool
+        self.handle = handle
+        self.err_on_empty = err_on_empty
+
+    def __call__(self, st: str) -> None:
+        if st == "waiting":
+            self.try_next()
+
+    def try_next(self) -> None:
+        pool = self.pool
+        handle = self.handle
+        if not handle.is_in_state("waiting"):
+            return
+        # The first try runs on the next loop turn; the pool may have
+        # started stopping (or failed) in between — fail now rather
+        # than queueing a waiter nothing will ever feed (companion to
+        # the stopping-state waiter drain).
+        if pool.is_in_state("stopping") or pool.is_in_state("stopped"):
+            handle.fail(mod_errors.PoolStoppingError(pool))
+            return
+        if pool.is_in_state("failed"):
+            handle.fail(mod_errors.PoolFailedError(
+                pool, pool.p_last_error))
+            return
+        # Idle connections sitting around?  Take one.  Entries may be
+        # stale ('stateChanged' is async): just unlink and skip them;
+        # the slot dispatcher copes (lib/pool.js:934-951).
+        idleq = pool.p_idleq
+        while len(idleq) > 0:
+            fsm = idleq.shift()
+            fsm.p_idleq_node = None
+            if not fsm.is_in_state("idle"):
+                continue
+            handle.try_(fsm)
+            return
+
+        if self.err_on_empty and pool.p_resolver.count() < 1:
+            handle.fail(mod_errors.NoBackendsError(
+                pool, pool.p_resolver.get_last_error()))
+
+        pool.p_waiters.push(handle)
+        pool._hwm_counter("max-claim-queue", len(pool.p_waiters))
+        pool._incr_counter("queued-claim")
+        pool.rebalance()
+
+
 class _CancelStub:
     """claim() return value when the pool is stopping/failed: supports
     only .cancel() (lib/pool.js:895-897)."""
@@ -732,50 +787,11 @@ class ConnectionPool(FSM):
         stack = mod_utils.maybe_capture_stack_trace()
         handle = ClaimHandle.fast(self, stack, cb, self.p_claim_log,
                                   timeout, self._loop)
-
-        def try_next() -> None:
-            if not handle.is_in_state("waiting"):
-                return
-            # The first try runs on the next loop turn; the pool may
-            # have started stopping (or failed) in between — fail now
-            # rather than queueing a waiter nothing will ever feed
-            # (companion to the stopping-state waiter drain).
-            if self.is_in_state("stopping") or self.is_in_state("stopped"):
-                handle.fail(mod_errors.PoolStoppingError(self))
-                return
-            if self.is_in_state("failed"):
-                handle.fail(mod_errors.PoolFailedError(
-                    self, self.p_last_error))
-                return
-            # Idle connections sitting around?  Take one.  Entries may be
-            # stale ('stateChanged' is async): just unlink and skip them;
-            # the slot dispatcher copes (lib/pool.js:934-951).
-            while len(self.p_idleq) > 0:
-                fsm = self.p_idleq.shift()
-                fsm.p_idleq_node = None
-                if not fsm.is_in_state("idle"):
-                    continue
-                handle.try_(fsm)
-                return
-
-            if err_on_empty and self.p_resolver.count() < 1:
-                handle.fail(mod_errors.NoBackendsError(
-                    self, self.p_resolver.get_last_error()))
-
-            self.p_waiters.push(handle)
-            self._hwm_counter("max-claim-queue", len(self.p_waiters))
-            self._incr_counter("queued-claim")
-            self.rebalance()
-
-        def waiting_listener(st: str) -> None:
-            if st == "waiting":
-                try_next()
-
         # The handle's construction already queued its async
-        # stateChanged('waiting'); the listener below receives it on the
-        # next loop turn and runs the first try_next then — claim() never
+        # stateChanged('waiting'); the ticket receives it on the next
+        # loop turn and runs the first try_next then — claim() never
         # fires the callback synchronously (lib/pool.js:922-968).
-        handle.on("stateChanged", waiting_listener)
+        handle.on("stateChanged", _ClaimTicket(self, handle, err_on_empty))
         return handle
 
     def _claim_shortcircuit(self, cb: Callable, err: BaseException):
