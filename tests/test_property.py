@@ -51,7 +51,7 @@ def test_pool_survives_random_interleavings(actions):
         loop.close()
 
 
-async def _scenario(loop, actions):
+async def _scenario(loop, actions, with_checker=False):
     conns = []
     resolver = DummyResolver()
     rfsm = ResolverFSM(resolver, {"loop": loop})
@@ -70,7 +70,7 @@ async def _scenario(loop, actions):
         c.destroy = destroy
         return c
 
-    pool = ConnectionPool({
+    pool_opts = {
         "domain": "prop.test",
         "constructor": ctor,
         "recovery": RECOVERY,
@@ -78,7 +78,17 @@ async def _scenario(loop, actions):
         "maximum": 4,
         "resolver": rfsm,
         "loop": loop,
-    })
+    }
+    if with_checker:
+        checked = []
+
+        def checker(hdl, conn):
+            checked.append(conn)
+            hdl.release()
+
+        pool_opts["checkTimeout"] = 150
+        pool_opts["checker"] = checker
+    pool = ConnectionPool(pool_opts)
     rfsm.start()
 
     backends = set()
@@ -353,3 +363,20 @@ async def _cset_scenario(loop, actions):
         for k, f in cset.cs_fsm.items()}
     # everything advertised was eventually removed
     assert not advertised
+
+
+# ---------------------------------------------------------------------------
+# pool fuzzing with the ping health-checker enabled: the checker's
+# internal claims (infinite timeout, initq-riding) interleave with
+# everything else
+
+@settings(max_examples=120, deadline=None,
+          suppress_health_check=[HealthCheck.too_slow])
+@given(actions=ACTIONS)
+def test_pool_with_checker_survives_interleavings(actions):
+    loop = VirtualLoop()
+    try:
+        loop.run_until_complete(
+            _scenario(loop, actions, with_checker=True))
+    finally:
+        loop.close()
