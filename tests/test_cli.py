@@ -66,3 +66,88 @@ def test_dns_resolution_against_mock(capsys):
     rc = main(["-r", "127.0.0.1@1", "-t", "200ms", "-p", "80",
                "doesnot.exist.test"])
     assert rc == 1  # resolver failed -> exit 1 (reference cbrFailed)
+
+
+def test_cli_dns_end_to_end_subprocess():
+    """bin/cbresolve resolves through a real mock DNS server."""
+    import asyncio
+    import os
+    import subprocess
+    import sys
+
+    ROOT = os.path.dirname(os.path.dirname(os.path.abspath(__file__)))
+
+    async def serve():
+        from cueball_amd.testing import MockDnsServer
+        srv = MockDnsServer()
+        await srv.start()
+        srv.add_srv("_web._tcp.shop.test", "s1.shop.test", 8443, ttl=60)
+        srv.add_a("s1.shop.test", "10.9.8.7", ttl=60)
+        proc = await asyncio.create_subprocess_exec(
+            sys.executable, os.path.join(ROOT, "bin", "cbresolve"),
+            "-r", srv.resolver_address, "-s", "_web._tcp",
+            "-t", "5s", "shop.test",
+            stdout=asyncio.subprocess.PIPE,
+            stderr=asyncio.subprocess.PIPE, cwd=ROOT)
+        out, err = await asyncio.wait_for(proc.communicate(), timeout=60)
+        srv.stop()
+        return proc.returncode, out.decode(), err.decode()
+
+    rc, out, err = asyncio.new_event_loop().run_until_complete(serve())
+    assert rc == 0, (out, err)
+    assert "10.9.8.7" in out
+    assert "8443" in out
+
+
+def test_cli_kang_listener():
+    """cbresolve -k serves /kang/snapshot while following."""
+    import asyncio
+    import json as mod_json
+    import os
+    import socket
+    import subprocess
+    import sys
+    import urllib.request
+
+    ROOT = os.path.dirname(os.path.dirname(os.path.abspath(__file__)))
+
+    s = socket.socket()
+    s.bind(("127.0.0.1", 0))
+    kang_port = s.getsockname()[1]
+    s.close()
+
+    async def serve():
+        from cueball_amd.testing import MockDnsServer
+        srv = MockDnsServer()
+        await srv.start()
+        srv.add_srv("_web._tcp.shop.test", "s1.shop.test", 8443, ttl=60)
+        srv.add_a("s1.shop.test", "10.9.8.7", ttl=60)
+        proc = await asyncio.create_subprocess_exec(
+            sys.executable, os.path.join(ROOT, "bin", "cbresolve"),
+            "-r", srv.resolver_address, "-s", "_web._tcp",
+            "-f", "-k", str(kang_port), "shop.test",
+            stdout=asyncio.subprocess.PIPE,
+            stderr=asyncio.subprocess.PIPE, cwd=ROOT)
+        try:
+            # wait for the follow-mode process to resolve + serve kang
+            data = None
+            for _ in range(100):
+                await asyncio.sleep(0.1)
+                try:
+                    with urllib.request.urlopen(
+                            "http://127.0.0.1:%d/kang/snapshot"
+                            % kang_port, timeout=2) as r:
+                        data = mod_json.loads(r.read())
+                    if data.get("dns_res"):
+                        break
+                except OSError:
+                    continue
+            assert data is not None
+            assert data["service"]["name"] == "cueball"
+            assert len(data["dns_res"]) >= 1
+        finally:
+            proc.terminate()
+            await proc.wait()
+            srv.stop()
+
+    asyncio.new_event_loop().run_until_complete(serve())
