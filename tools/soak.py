@@ -11,8 +11,12 @@ visible.  Used for the endurance numbers recorded in BASELINE.md.
 import argparse
 import asyncio
 import gc
+import os
 import resource
+import sys
 import time
+
+sys.path.insert(0, os.path.dirname(os.path.dirname(os.path.abspath(__file__))))
 
 from cueball_amd.connection import tcp_constructor
 from cueball_amd.pool import ConnectionPool
