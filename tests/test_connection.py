@@ -143,3 +143,52 @@ def test_constructor_factory():
         srv.close()
 
     run(body())
+
+
+def test_pool_claim_over_ipv6_loopback():
+    """End-to-end claim -> echo -> release over a real ::1 TCP socket
+    (the reference supports v6 backends throughout; srvKey normalizes
+    v6 text forms, lib/resolver.js:1157-1171)."""
+    import asyncio
+
+    from cueball_amd.connection import tcp_constructor
+    from cueball_amd.pool import ConnectionPool
+    from cueball_amd.resolver import StaticIpResolver
+
+    async def body():
+        async def on_conn(reader, writer):
+            data = await reader.read(64)
+            writer.write(b"echo:" + data)
+            await writer.drain()
+            writer.close()
+
+        server = await asyncio.start_server(on_conn, "::1", 0)
+        port = server.sockets[0].getsockname()[1]
+
+        res = StaticIpResolver({"backends": [
+            {"address": "::1", "port": port}]})
+        pool = ConnectionPool({
+            "domain": "v6.test", "resolver": res,
+            "constructor": tcp_constructor(),
+            "spares": 1, "maximum": 2,
+            "recovery": {"default": {"timeout": 2000, "retries": 2,
+                                     "delay": 100, "maxDelay": 500}},
+        })
+        res.start()
+        handle, conn = await asyncio.wait_for(pool.claim_async(), 10)
+        got = asyncio.get_running_loop().create_future()
+        conn.on("data", lambda d: (not got.done()) and got.set_result(d))
+        conn.write(b"ping6")
+        data = await asyncio.wait_for(got, 5)
+        assert data == b"echo:ping6"
+        handle.close()
+
+        fut = asyncio.get_running_loop().create_future()
+        pool.on("stateChanged", lambda st: st == "stopped"
+                and not fut.done() and fut.set_result(None))
+        pool.stop()
+        await asyncio.wait_for(fut, 10)
+        server.close()
+        await server.wait_closed()
+
+    asyncio.new_event_loop().run_until_complete(body())
