@@ -93,7 +93,11 @@ def test_cli_dns_end_to_end_subprocess():
         srv.stop()
         return proc.returncode, out.decode(), err.decode()
 
-    rc, out, err = asyncio.new_event_loop().run_until_complete(serve())
+    loop = asyncio.new_event_loop()
+    try:
+        rc, out, err = loop.run_until_complete(serve())
+    finally:
+        loop.close()
     assert rc == 0, (out, err)
     assert "10.9.8.7" in out
     assert "8443" in out
@@ -150,4 +154,8 @@ def test_cli_kang_listener():
             await proc.wait()
             srv.stop()
 
-    asyncio.new_event_loop().run_until_complete(serve())
+    loop = asyncio.new_event_loop()
+    try:
+        loop.run_until_complete(serve())
+    finally:
+        loop.close()

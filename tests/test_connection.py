@@ -191,4 +191,8 @@ def test_pool_claim_over_ipv6_loopback():
         server.close()
         await server.wait_closed()
 
-    asyncio.new_event_loop().run_until_complete(body())
+    loop = asyncio.new_event_loop()
+    try:
+        loop.run_until_complete(body())
+    finally:
+        loop.close()

@@ -18,7 +18,11 @@ from cueball_amd.testing import MockDnsServer
 
 
 def run(coro):
-    return asyncio.new_event_loop().run_until_complete(coro)
+    loop = asyncio.new_event_loop()
+    try:
+        return loop.run_until_complete(coro)
+    finally:
+        loop.close()
 
 
 def test_wire_roundtrip():
