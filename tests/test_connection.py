@@ -140,7 +140,9 @@ def test_constructor_factory():
         conn.ref()
         conn.unref()
         conn.destroy()
+        await asyncio.sleep(0.05)  # let the server handler see EOF
         srv.close()
+        await srv.wait_closed()
 
     run(body())
 
