@@ -464,3 +464,39 @@ def test_streaming_write_requires_flag():
     req = HttpRequest("POST", "/x", host="h")
     with pytest.raises(RuntimeError):
         req.write(b"nope")
+
+
+def test_agent_with_custom_resolver():
+    """create_pool(host, {resolver}) routes requests for an arbitrary
+    hostname through a user-supplied resolver object
+    (reference test/agent.test.js:214-265)."""
+    async def body():
+        from cueball_amd.resolver import StaticIpResolver
+        srv1 = MockHttpServer()
+        srv2 = MockHttpServer()
+        await srv1.start()
+        await srv2.start()
+        res = StaticIpResolver({"backends": [
+            {"address": "127.0.0.1", "port": srv1.port},
+            {"address": "127.0.0.1", "port": srv2.port},
+        ]})
+        agent = HttpAgent({
+            "recovery": RECOVERY,
+            "spares": 2,
+            "maximum": 4,
+        })
+        assert agent.get_pool("foobar") is None
+        agent.create_pool("foobar", {"resolver": res})
+        assert agent.get_pool("foobar") is not None
+        assert res.is_in_state("stopped")
+        res.start()
+        resp = await asyncio.wait_for(
+            agent.request_async("foobar", "GET", "/test4"), 10)
+        assert resp.status_code == 200
+        # the request landed on one of the two resolver-listed servers
+        assert srv1.request_count + srv2.request_count >= 1
+        await stop_agent(agent)
+        srv1.stop()
+        srv2.stop()
+
+    run(body())
