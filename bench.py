@@ -23,7 +23,6 @@ static1, dns, cset, codel, agent — select with --config.
 import argparse
 import asyncio
 import json
-import math
 import os
 import statistics
 import sys
