@@ -241,6 +241,14 @@ class ConnectionPool(FSM):
         self.p_lpf = FIRFilter(LP_TAPS)
         self.p_last_rebal_clamped = False
         self.p_rate_delay_timer = None
+        # Census signature of the last no-op rebalance: plan_rebalance
+        # is a pure function of (key order, per-key counts, dead set,
+        # target, max), so if none of those changed since a rebalance
+        # that planned nothing, the plan is empty again and we can skip
+        # building it.  Claim-heavy steady state schedules a rebalance
+        # every loop turn (each idle->busy claim does), making this the
+        # hot no-op path.
+        self.p_noop_census: Optional[tuple] = None
 
         self.p_idleq = Queue()
         self.p_initq = Queue()
@@ -537,10 +545,11 @@ class ConnectionPool(FSM):
         self.p_rebal_scheduled = False
 
         total = 0
-        conns: Dict[str, List[ConnectionSlotFSM]] = {}
+        counts = []
         for k in self.p_keys:
-            conns[k] = list(self.p_connections.get(k, ()))
-            total += len(conns[k])
+            n = len(self.p_connections.get(k, ()))
+            counts.append(n)
+            total += n
         spares = len(self.p_idleq) + len(self.p_initq) - len(self.p_waiters)
         if spares < 0:
             spares = 0
@@ -564,7 +573,20 @@ class ConnectionPool(FSM):
         if target > self.p_max:
             target = self.p_max
 
+        census = (target, tuple(counts), tuple(self.p_keys),
+                  tuple(sorted(self.p_dead)))
+        if census == self.p_noop_census:
+            self.p_in_rebalance = False
+            self.p_last_rebalance = mod_time.time()
+            return
+
+        conns: Dict[str, List[ConnectionSlotFSM]] = {}
+        for k in self.p_keys:
+            conns[k] = list(self.p_connections.get(k, ()))
+
         plan = mod_utils.plan_rebalance(conns, self.p_dead, target, self.p_max)
+        self.p_noop_census = None if (plan["remove"] or plan["add"]) \
+            else census
 
         if plan["remove"] or plan["add"]:
             self.p_log.trace(
