@@ -556,7 +556,7 @@ def main():
             "higher_is_better": True,
             "scaling": "weak",
             "vs_baseline": None,
-            "dtype": "fp64",
+            "dtype": "n/a",
             "data": "synthetic",
             "config": {
                 "model": "ConnectionPool claim/release (config: %s)"

@@ -164,6 +164,19 @@ class Agent(EventEmitter):
                         import socket as mod_socket
                         sock.setsockopt(mod_socket.SOL_SOCKET,
                                         mod_socket.SO_KEEPALIVE, 1)
+                        # node's setKeepAlive(initialDelay) also sets the
+                        # idle time before the first probe; map the ms
+                        # option onto TCP_KEEPIDLE (whole seconds, >=1).
+                        idle_s = max(1, int(agent.tcp_ka_delay / 1000))
+                        for opt in ("TCP_KEEPIDLE", "TCP_KEEPALIVE"):
+                            if hasattr(mod_socket, opt):
+                                try:
+                                    sock.setsockopt(
+                                        mod_socket.IPPROTO_TCP,
+                                        getattr(mod_socket, opt), idle_s)
+                                except OSError:
+                                    pass
+                                break
                 conn.on("connect", enable_ka)
             return conn
 

@@ -34,8 +34,12 @@ class ControlledDelay:
         self.cd_drop_next = 0.0
         self.cd_count = 0
         self.cd_dropping = False
-        self.cd_last_empty = 0.0
         self._loop = get_loop(loop)
+        # Treat the queue as having just been empty at construction so a
+        # freshly started pool uses the healthy 10x-target bound from
+        # get_max_idle(), matching the reference where the initial
+        # undefined compare evaluates false (lib/codel.js:109-118).
+        self.cd_last_empty = self._now()
 
     def _now(self) -> float:
         return self._loop.time() * 1000.0

@@ -126,7 +126,11 @@ class SocketMgrFSM(FSM):
         self.sm_max_delay = recov.get("maxDelay") or math.inf
         self.sm_timeout = recov["timeout"]
         self.sm_max_timeout = recov.get("maxTimeout") or math.inf
-        self.sm_delay_spread = recov.get("delaySpread") or 0.2
+        # Default only when absent: an explicit delaySpread of 0.0 is a
+        # deliberate "no spread" setting and must be preserved
+        # (reference defaults only on undefined).
+        spread = recov.get("delaySpread")
+        self.sm_delay_spread = 0.2 if spread is None else spread
 
         if self.sm_monitor is True:
             mult = 1 << int(self.sm_retries)

@@ -231,7 +231,9 @@ class ConnectionSet(FSM):
                         one_done()
                 return cb
 
-            fsm.on("stateChanged", make_cb())
+            # Scoped: auto-removed when the set leaves this state (see
+            # the matching fix in pool.state_stopping_backends).
+            S.on(fsm, "stateChanged", make_cb())
             fsm.set_unwanted()
             for ck in cks:
                 # async: avoid FSM loops when .stop() was called from an

@@ -16,6 +16,7 @@ from __future__ import annotations
 
 import asyncio
 import random
+import weakref
 from typing import Any, Callable, Dict, List, Optional
 
 from . import dns_wire
@@ -103,13 +104,18 @@ class DnsClient:
 
     def __init__(self, concurrency: int = 3) -> None:
         self.concurrency = concurrency
-        self._sems: Dict[int, asyncio.Semaphore] = {}
+        # Keyed by the loop object itself through a WeakKeyDictionary so
+        # that (a) destroyed loops release their semaphore and (b) id()
+        # reuse can never hand a dead loop's semaphore to a new loop.
+        self._sems: ("weakref.WeakKeyDictionary"
+                     "[asyncio.AbstractEventLoop, asyncio.Semaphore]") = (
+            weakref.WeakKeyDictionary())
 
     def _sem(self, loop: asyncio.AbstractEventLoop) -> asyncio.Semaphore:
-        sem = self._sems.get(id(loop))
+        sem = self._sems.get(loop)
         if sem is None:
             sem = asyncio.Semaphore(self.concurrency)
-            self._sems[id(loop)] = sem
+            self._sems[loop] = sem
         return sem
 
     # -- callback adapter used by the resolver FSM ----------------------
@@ -245,7 +251,14 @@ class DnsClient:
                     timeout=max(0.001, deadline - loop.time()))
             except asyncio.TimeoutError:
                 raise TimeoutError_(domain, server) from None
-            return dns_wire.decode_message(data)
+            msg = dns_wire.decode_message(data)
+            # Same spoof/desync check as the UDP path: the response on the
+            # truncation-fallback TCP stream must echo our query id.
+            if msg.id != qid:
+                raise DnsError(
+                    "DNS server %s returned mismatched query id %d "
+                    "(expected %d) for %s" % (server, msg.id, qid, domain))
+            return msg
         finally:
             writer.close()
 

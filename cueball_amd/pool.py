@@ -486,7 +486,11 @@ class ConnectionPool(FSM):
                             one_done()
                     return cb
 
-                fsm.on("stateChanged", make_cb())
+                # Scoped: auto-removed from the slot FSM when the pool
+                # leaves stopping.backends, so repeated stop cycles (or
+                # slots that outlive the pool) cannot accumulate
+                # listeners (round-1 review finding).
+                S.on(fsm, "stateChanged", make_cb())
 
     def state_stopped(self, S: StateScope) -> None:
         S.valid_transitions([])

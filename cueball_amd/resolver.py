@@ -257,7 +257,9 @@ class DNSResolverFSM(FSM):
                 "timeout": r["timeout"],
                 "minDelay": r["delay"],
                 "delay": r["delay"],
-                "delaySpread": r.get("delaySpread") or 0.2,
+                # default only when absent; explicit 0.0 spread is kept
+                "delaySpread": (0.2 if r.get("delaySpread") is None
+                                else r["delaySpread"]),
                 "maxDelay": r.get("maxDelay") or math.inf,
             }
 
