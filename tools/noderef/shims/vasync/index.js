@@ -1,0 +1,75 @@
+/*
+ * vasync shim: forEachParallel / forEachPipeline, the only entry
+ * points node-cueball uses (teardown fan-in paths).
+ */
+
+'use strict';
+
+function forEachParallel(opts, cb) {
+	const inputs = opts.inputs;
+	const func = opts.func;
+	const results = {
+		operations: [],
+		successes: [],
+		ndone: 0,
+		nerrors: 0
+	};
+	var firstErr = null;
+	if (inputs.length === 0) {
+		setImmediate(function () { cb(null, results); });
+		return (results);
+	}
+	var remaining = inputs.length;
+	inputs.forEach(function (input, i) {
+		const op = { input: input, status: 'pending',
+		    err: null, result: null };
+		results.operations[i] = op;
+		func(input, function (err, res) {
+			op.status = err ? 'fail' : 'ok';
+			op.err = err || null;
+			op.result = res;
+			results.ndone++;
+			if (err) {
+				results.nerrors++;
+				if (firstErr === null)
+					firstErr = err;
+			} else {
+				results.successes.push(res);
+			}
+			if (--remaining === 0)
+				cb(firstErr, results);
+		});
+	});
+	return (results);
+}
+
+function forEachPipeline(opts, cb) {
+	const inputs = opts.inputs.slice();
+	const func = opts.func;
+	const results = {
+		operations: [], successes: [], ndone: 0, nerrors: 0
+	};
+	var i = 0;
+	function next(err) {
+		if (err || i >= inputs.length) {
+			cb(err || null, results);
+			return;
+		}
+		const input = inputs[i++];
+		func(input, function (e, res) {
+			results.ndone++;
+			if (e)
+				results.nerrors++;
+			else
+				results.successes.push(res);
+			next(e);
+		});
+	}
+	next(null);
+	return (results);
+}
+
+module.exports = {
+	forEachParallel: forEachParallel,
+	forEachPipeline: forEachPipeline
+};
