@@ -896,36 +896,80 @@ FSM_flush_state_changed(PyObject *self_, PyObject *noargs)
 /* Batched flush: all FSMs that queue a stateChanged in the same loop
  * turn share one call_soon.  Relative order is preserved (FSMs drain in
  * first-queue order; each FSM drains its whole queue, exactly as the
- * per-FSM call_soon did).  A batch already being drained is never
- * appended to: later queuers start a fresh batch + call_soon, keeping
- * the ordering identical to the unbatched scheme. */
+ * per-FSM call_soon did).  While the batch is draining it stays
+ * registered, so FSMs that queue *during* the drain (the serial
+ * claim-chain case: a flush completes one claim cycle, whose callback
+ * immediately starts the next claim) are appended to the same batch
+ * and drained in the same loop callback.  That collapses long
+ * callback-chained sequences into one event-loop turn instead of one
+ * turn (epoll_wait + ready-queue bookkeeping) per link.  The drain is
+ * capped per callback so a self-sustaining chain cannot starve IO; the
+ * remainder is rescheduled with a fresh call_soon. */
+#define FLUSH_DRAIN_CAP 256
+
 typedef struct {
     PyObject_HEAD
     PyObject *fb_loop;
     PyObject *fb_list;   /* list of FSMOb */
+    Py_ssize_t fb_pos;   /* next index to drain */
 } FlushBatch;
 
 extern PyTypeObject FlushBatchType;
+
+static void
+flushbatch_detach(FlushBatch *self)
+{
+    PyObject *cur = PyDict_GetItemWithError(g_flush_batches, self->fb_loop);
+    if (cur == (PyObject *)self)
+        (void)PyDict_DelItem(g_flush_batches, self->fb_loop);
+    else if (cur == NULL)
+        PyErr_Clear();
+}
 
 PyObject *
 FlushBatch_call(PyObject *self_, PyObject *args, PyObject *kwds)
 {
     FlushBatch *self = (FlushBatch *)self_;
     (void)args; (void)kwds;
-    /* detach ourselves: new queuers start a new batch */
-    PyObject *cur = PyDict_GetItemWithError(g_flush_batches, self->fb_loop);
-    if (cur == (PyObject *)self) {
-        if (PyDict_DelItem(g_flush_batches, self->fb_loop) < 0)
+    int rounds = 0;
+    while (self->fb_pos < PyList_GET_SIZE(self->fb_list) &&
+           rounds < FLUSH_DRAIN_CAP) {
+        FSMOb *fsm = (FSMOb *)PyList_GET_ITEM(self->fb_list,
+                                              self->fb_pos);
+        self->fb_pos++;
+        rounds++;
+        if (fsm_flush_core(fsm) < 0) {
+            /* un-wedge the FSMs we did not get to: their next
+             * emission must be able to schedule a flush again */
+            for (Py_ssize_t i = self->fb_pos;
+                 i < PyList_GET_SIZE(self->fb_list); i++) {
+                FSMOb *f = (FSMOb *)PyList_GET_ITEM(self->fb_list, i);
+                f->f_emit_scheduled = 0;
+            }
+            flushbatch_detach(self);
             return NULL;
-    } else if (cur == NULL && PyErr_Occurred()) {
-        return NULL;
+        }
     }
-    Py_ssize_t n = PyList_GET_SIZE(self->fb_list);
-    for (Py_ssize_t i = 0; i < n; i++) {
-        FSMOb *fsm = (FSMOb *)PyList_GET_ITEM(self->fb_list, i);
-        if (fsm_flush_core(fsm) < 0)
+    if (self->fb_pos < PyList_GET_SIZE(self->fb_list)) {
+        /* cap hit: yield to the loop, keep the batch registered so
+         * new queuers keep appending, and finish on the next turn.
+         * Compact the drained prefix first — under sustained load the
+         * batch never empties and would otherwise grow without bound. */
+        if (PyList_SetSlice(self->fb_list, 0, self->fb_pos, NULL) < 0) {
+            flushbatch_detach(self);
             return NULL;
+        }
+        self->fb_pos = 0;
+        PyObject *h = PyObject_CallMethodObjArgs(
+            self->fb_loop, s_call_soon, self_, NULL);
+        if (h == NULL) {
+            flushbatch_detach(self);
+            return NULL;
+        }
+        Py_DECREF(h);
+        Py_RETURN_NONE;
     }
+    flushbatch_detach(self);
     Py_RETURN_NONE;
 }
 
@@ -987,6 +1031,7 @@ fsm_queue_state_changed(FSMOb *self, PyObject *state)
             return -1;
         Py_INCREF(self->f_loop);
         fb->fb_loop = self->f_loop;
+        fb->fb_pos = 0;
         fb->fb_list = PyList_New(0);
         PyObject_GC_Track((PyObject *)fb);
         if (fb->fb_list == NULL) {

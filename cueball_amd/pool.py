@@ -122,6 +122,7 @@ class _ClaimTicket:
             if not fsm.is_in_state("idle"):
                 continue
             handle.try_(fsm)
+            pool._note_demand()
             return
 
         if self.err_on_empty and pool.p_resolver.count() < 1:
@@ -129,6 +130,7 @@ class _ClaimTicket:
                 pool, pool.p_resolver.get_last_error()))
 
         pool.p_waiters.push(handle)
+        pool._note_demand()
         pool._hwm_counter("max-claim-queue", len(pool.p_waiters))
         pool._incr_counter("queued-claim")
         pool.rebalance()
@@ -249,6 +251,19 @@ class ConnectionPool(FSM):
         # every loop turn (each idle->busy claim does), making this the
         # hot no-op path.
         self.p_noop_census: Optional[tuple] = None
+        # Peak demand (busy + extras) and peak busy seen since the last
+        # rebalance / LPF sample.  The reference sizes from an
+        # instantaneous sample taken when its setImmediate/timer fires
+        # (lib/pool.js:560-588, :251-263); with this runtime's batched
+        # event dispatch those callbacks tend to run at quiescent
+        # points between claim bursts, where busy == waiters == 0, so
+        # an instantaneous sample would chronically under-size the
+        # pool.  Tracking the high-water mark on the claim path sizes
+        # for the peak concurrent demand of the interval instead,
+        # which is the reference's intent.
+        self.p_demand_hwm = 0
+        self.p_busy_hwm = 0
+        self.p_total_conns = 0
 
         self.p_idleq = Queue()
         self.p_initq = Queue()
@@ -300,11 +315,34 @@ class ConnectionPool(FSM):
         if self.p_counters.get(counter, -1) < val:
             self.p_counters[counter] = val
 
+    def _note_demand(self) -> None:
+        """O(1) demand sample on the claim path: fold the current
+        busy/extras into the high-water marks consumed by _rebalance
+        and _lp_tick (see p_demand_hwm comment in __init__)."""
+        nw = len(self.p_waiters)
+        ni = len(self.p_initq)
+        spares = len(self.p_idleq) + ni - nw
+        if spares < 0:
+            spares = 0
+        busy = self.p_total_conns - spares
+        if busy < 0:
+            busy = 0
+        extras = nw - ni
+        if extras < 0:
+            extras = 0
+        if busy > self.p_busy_hwm:
+            self.p_busy_hwm = busy
+        if busy + extras > self.p_demand_hwm:
+            self.p_demand_hwm = busy + extras
+
     # -- LPF anti-shrink sampling (lib/pool.js:251-263) ------------------
     def _lp_tick(self) -> None:
         conns = sum(len(v) for v in self.p_connections.values())
         spares = len(self.p_idleq) + len(self.p_initq)
         busy = conns - spares
+        if busy < self.p_busy_hwm:
+            busy = self.p_busy_hwm
+        self.p_busy_hwm = conns - spares if conns > spares else 0
         self.p_lpf.put(busy + self.p_spares)
         # piggyback live gauges on the 5 Hz sample (beyond the
         # reference, which exposes state through kang only)
@@ -564,7 +602,14 @@ class ConnectionPool(FSM):
         if extras < 0:
             extras = 0
 
-        target = busy + extras + self.p_spares
+        # Size for the peak demand of the interval, not just this
+        # instant (see p_demand_hwm comment in __init__).
+        demand = busy + extras
+        if demand < self.p_demand_hwm:
+            demand = self.p_demand_hwm
+        self.p_demand_hwm = busy + extras
+
+        target = demand + self.p_spares
 
         # Anti-shrink clamp from the low-pass filter (lib/pool.js:579-588)
         lpf_min = math.ceil(self.p_lpf.get())
@@ -671,6 +716,7 @@ class ConnectionPool(FSM):
             "loop": self._loop,
         })
         self.p_connections.setdefault(key, []).append(fsm)
+        self.p_total_conns += 1
 
         fsm.p_initq_node = self.p_initq.push(fsm)
         fsm.p_idleq_node = None
@@ -737,6 +783,7 @@ class ConnectionPool(FSM):
             conns = self.p_connections.get(key)
             if conns is not None:
                 conns.remove(fsm)
+                self.p_total_conns -= 1
                 if not conns:
                     del self.p_connections[key]
             self.emit("closedBackend", key, fsm)

@@ -167,6 +167,27 @@ ClaimDriver.prototype.claimOnce = function () {
 	});
 };
 
+/*
+ * Multi-worker barrier: when forked by bench_ref_multi.js
+ * (BENCH_WORKER=1), wait for every worker to finish warmup before the
+ * timed region starts — the analog of bench.py's gloo barrier.
+ */
+function barrier(cb) {
+	if (process.env.BENCH_WORKER !== '1' ||
+	    process.send === undefined) {
+		cb();
+		return;
+	}
+	process.once('message', function onMsg(m) {
+		if (m && m.type === 'go') {
+			cb();
+			return;
+		}
+		process.once('message', onMsg);
+	});
+	process.send({ type: 'ready' });
+}
+
 function runSteps(driver, args, results, done) {
 	var i = 0;
 	function warm() {
@@ -174,20 +195,24 @@ function runSteps(driver, args, results, done) {
 			driver.runStep(args.claimsPerStep, warm);
 			return;
 		}
-		driver.latencies.length = 0;
-		const t0 = process.hrtime();
-		var j = 0;
-		function step() {
-			if (j++ < args.steps) {
-				driver.runStep(args.claimsPerStep, step);
-				return;
+		barrier(function () {
+			driver.latencies.length = 0;
+			const t0 = process.hrtime();
+			var j = 0;
+			function step() {
+				if (j++ < args.steps) {
+					driver.runStep(args.claimsPerStep,
+					    step);
+					return;
+				}
+				const d = process.hrtime(t0);
+				results.elapsed = d[0] + d[1] / 1e9;
+				results.ops = args.steps *
+				    args.claimsPerStep;
+				done();
 			}
-			const d = process.hrtime(t0);
-			results.elapsed = d[0] + d[1] / 1e9;
-			results.ops = args.steps * args.claimsPerStep;
-			done();
-		}
-		step();
+			step();
+		});
 	}
 	warm();
 }
@@ -717,6 +742,16 @@ function main() {
 	if (fn === undefined)
 		throw (new Error('unknown config ' + args.config));
 	fn(args, results, function () {
+		if (process.env.BENCH_WORKER === '1' &&
+		    process.send !== undefined) {
+			process.send({ type: 'result',
+			    elapsed: results.elapsed, ops: results.ops,
+			    lat_p50: results.lat_p50,
+			    lat_p99: results.lat_p99,
+			    shed: results.shed });
+			setTimeout(function () { process.exit(0); }, 200);
+			return;
+		}
 		const value = results.ops / results.elapsed;
 		const out = {
 			metric: 'pool claims/sec (8-backend synthetic ' +
