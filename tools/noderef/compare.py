@@ -94,6 +94,21 @@ def main():
         cset_steps = 12
         codel_claims = 20000
 
+    # Hosts without /root/reference (e.g. the benchmark box) get the
+    # vendored copy of the reference sources (see README.md here).
+    if not os.environ.get("CUEBALL_REF") and \
+            not os.path.isdir("/root/reference/lib"):
+        import tarfile
+        import tempfile
+        tgz = os.path.join(REPO, "tools", "noderef",
+                           "reference-cueball-2.10.3.tgz")
+        tmp = tempfile.mkdtemp(prefix="noderef-")
+        with tarfile.open(tgz) as tf:
+            tf.extractall(tmp)
+        os.environ["CUEBALL_REF"] = os.path.join(tmp, "reference")
+        print("extracted reference to %s" % os.environ["CUEBALL_REF"],
+              flush=True)
+
     results = {"host": os.uname().nodename,
                "node": None, "python": sys.version.split()[0],
                "scenarios": {}, "scaling": {}}
