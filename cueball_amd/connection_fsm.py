@@ -616,6 +616,18 @@ class ConnectionSlotFSM(FSM):
         self.csf_wanted = True
         self.csf_handle: Optional[ClaimHandle] = None
         self.csf_prev_handle: Optional[ClaimHandle] = None
+        # busy/idle hot-cycle state (see _busy_* / _idle_* helpers):
+        # scopes of the current busy/idle states plus the
+        # event-observed smgr state, with the callbacks bound once here
+        # instead of allocating closures on every claim.
+        self.csf_observed = "connected"
+        self.csf_busy_scope: Optional[StateScope] = None
+        self.csf_idle_scope: Optional[StateScope] = None
+        self._b_busy_smgr = self._busy_smgr_cb
+        self._b_busy_hdl = self._busy_hdl_cb
+        self._b_idle_smgr = self._idle_smgr_cb
+        self._b_idle_unwanted = self._idle_on_unwanted
+        self._b_idle_ping = self._idle_ping_cb
         self.csf_monitor = bool(options["monitor"])
         self.csf_checker = options.get("checker")
         self.csf_check_timeout = options.get("checkTimeout")
@@ -737,6 +749,7 @@ class ConnectionSlotFSM(FSM):
         S.valid_transitions(["retrying", "connecting", "stopping", "stopped",
                              "busy"])
         smgr = self.csf_smgr
+        self.csf_idle_scope = S
 
         if self.csf_handle is not None:
             self.csf_prev_handle = self.csf_handle
@@ -759,94 +772,110 @@ class ConnectionSlotFSM(FSM):
                 smgr.sm_monitor = False
                 smgr.reset_backoff()
 
-        def on_unwanted() -> None:
-            if smgr.is_in_state("connected"):
-                S.goto_state("stopping")
-            elif smgr.is_in_state("error") or smgr.is_in_state("closed"):
-                # Divergence from the reference (bug fix, found by the
-                # property suite): a slot flagged unwanted whose socket
-                # connected and then died within the same loop spin
-                # reaches idle with the smgr already in error/closed.
-                # The reference's early-return would leave the slot
-                # wedged in idle with no listeners registered
-                # (lib/connection-fsm.js:1059-1062 registers nothing
-                # when !csf_wanted).  An unwanted slot with a dead
-                # socket is simply done.
-                S.goto_state("stopped")
-
-        def on_smgr_state(st: str) -> None:
-            if st == "error":
-                if self.csf_wanted:
-                    S.goto_state("retrying")
-                else:
-                    S.goto_state("stopped")
-            elif st == "closed":
-                if not self.csf_wanted:
-                    S.goto_state("stopped")
-                else:
-                    S.goto_state("connecting")
-            else:
-                raise FSMError(
-                    'Unhandled smgr state transition: connected => "%s"' % st)
-
         # register the smgr listener in every case — pending smgr events
         # must find a handler even when the slot is already unwanted
-        S.on(smgr, "stateChanged", on_smgr_state)
+        S.on(smgr, "stateChanged", self._b_idle_smgr)
 
         if not self.csf_wanted:
-            on_unwanted()
+            self._idle_on_unwanted()
             return
-        S.on(self, "unwanted", on_unwanted)
+        S.on(self, "unwanted", self._b_idle_unwanted)
 
         if self.csf_check_timeout is not None and \
                 self.csf_checker is not None:
-            S.timeout(self.csf_check_timeout,
-                      lambda: _do_ping_check(self, self.csf_checker))
+            S.timeout(self.csf_check_timeout, self._b_idle_ping)
+
+    def _idle_on_unwanted(self) -> None:
+        S = self.csf_idle_scope
+        smgr = self.csf_smgr
+        if smgr.is_in_state("connected"):
+            S.goto_state("stopping")
+        elif smgr.is_in_state("error") or smgr.is_in_state("closed"):
+            # Divergence from the reference (bug fix, found by the
+            # property suite): a slot flagged unwanted whose socket
+            # connected and then died within the same loop spin
+            # reaches idle with the smgr already in error/closed.
+            # The reference's early-return would leave the slot
+            # wedged in idle with no listeners registered
+            # (lib/connection-fsm.js:1059-1062 registers nothing
+            # when !csf_wanted).  An unwanted slot with a dead
+            # socket is simply done.
+            S.goto_state("stopped")
+
+    def _idle_smgr_cb(self, st: str) -> None:
+        S = self.csf_idle_scope
+        if st == "error":
+            if self.csf_wanted:
+                S.goto_state("retrying")
+            else:
+                S.goto_state("stopped")
+        elif st == "closed":
+            if not self.csf_wanted:
+                S.goto_state("stopped")
+            else:
+                S.goto_state("connecting")
+        else:
+            raise FSMError(
+                'Unhandled smgr state transition: connected => "%s"' % st)
+
+    def _idle_ping_cb(self) -> None:
+        _do_ping_check(self, self.csf_checker)
+
+    # The busy-state helpers are pre-bound methods (see __init__)
+    # rather than per-entry closures: the busy/idle cycle runs once per
+    # claim, and closure allocation was a measurable share of the claim
+    # path's memory traffic.  csf_busy_scope carries the state scope;
+    # a callback whose scope has been exited no-ops in goto_state, the
+    # same guard the closures had by capturing S.
+    def _busy_smgr_cb(self, st: str) -> None:
+        # Track the smgr state as *observed through events*: a
+        # transition that happened this same loop spin is still pending
+        # delivery, and the exit decision must match what we have
+        # actually seen (lib/connection-fsm.js:885-890, :1129-1196).
+        self.csf_observed = st
+
+    def _busy_on_release(self) -> None:
+        S = self.csf_busy_scope
+        st = self.csf_observed
+        if st == "connected":
+            if self.csf_wanted:
+                S.goto_state("idle")
+            else:
+                S.goto_state("stopping")
+        elif st == "closed":
+            if self.csf_wanted:
+                S.goto_state("connecting")
+            else:
+                S.goto_state("stopped")
+        elif st == "error":
+            S.goto_state("retrying")
+        else:
+            raise FSMError(
+                'Handle released while smgr was in unhandled state "%s"'
+                % self.csf_smgr.get_state())
+
+    def _busy_on_close(self) -> None:
+        S = self.csf_busy_scope
+        if self.csf_observed == "connected":
+            S.goto_state("killing")
+        else:
+            S.goto_state("retrying")
+
+    def _busy_hdl_cb(self, st: str) -> None:
+        if st == "released":
+            self._busy_on_release()
+        elif st == "closed":
+            self._busy_on_close()
 
     def state_busy(self, S: StateScope) -> None:
         S.valid_transitions(["idle", "stopping", "stopped", "retrying",
                              "killing", "connecting"])
         smgr = self.csf_smgr
         hdl = self.csf_handle
-        # Track the smgr state as *observed through events*: a transition
-        # that happened this same loop spin is still pending delivery, and
-        # the exit decision must match what we have actually seen
-        # (lib/connection-fsm.js:885-890, :1129-1196).
-        observed = {"st": "connected"}
-        S.on(smgr, "stateChanged", lambda st: observed.update(st=st))
-
-        def on_release() -> None:
-            st = observed["st"]
-            if st == "connected":
-                if self.csf_wanted:
-                    S.goto_state("idle")
-                else:
-                    S.goto_state("stopping")
-            elif st == "closed":
-                if self.csf_wanted:
-                    S.goto_state("connecting")
-                else:
-                    S.goto_state("stopped")
-            elif st == "error":
-                S.goto_state("retrying")
-            else:
-                raise FSMError(
-                    'Handle released while smgr was in unhandled state "%s"'
-                    % smgr.get_state())
-
-        def on_close() -> None:
-            if observed["st"] == "connected":
-                S.goto_state("killing")
-            else:
-                S.goto_state("retrying")
-
-        def on_hdl_state(st: str) -> None:
-            if st == "released":
-                on_release()
-            elif st == "closed":
-                on_close()
-
-        S.on(hdl, "stateChanged", on_hdl_state)
+        self.csf_observed = "connected"
+        self.csf_busy_scope = S
+        S.on(smgr, "stateChanged", self._b_busy_smgr)
+        S.on(hdl, "stateChanged", self._b_busy_hdl)
 
         # The smgr may have left 'connected' before we got here; if we
         # lost the race, treat it like our handle was released.
@@ -855,7 +884,7 @@ class ConnectionSlotFSM(FSM):
         else:
             hdl.reject()
             self.csf_handle = None
-            on_release()
+            self._busy_on_release()
 
     def state_killing(self, S: StateScope) -> None:
         S.valid_transitions(["retrying"])
