@@ -745,9 +745,11 @@ class ConnectionSlotFSM(FSM):
         S.on(self, "unwanted", on_unwanted)
         smgr.retry()
 
+    _IDLE_VALID = ("retrying", "connecting", "stopping", "stopped",
+                   "busy")
+
     def state_idle(self, S: StateScope) -> None:
-        S.valid_transitions(["retrying", "connecting", "stopping", "stopped",
-                             "busy"])
+        S.valid_transitions(self._IDLE_VALID)
         smgr = self.csf_smgr
         self.csf_idle_scope = S
 
@@ -867,9 +869,11 @@ class ConnectionSlotFSM(FSM):
         elif st == "closed":
             self._busy_on_close()
 
+    _BUSY_VALID = ("idle", "stopping", "stopped", "retrying",
+                   "killing", "connecting")
+
     def state_busy(self, S: StateScope) -> None:
-        S.valid_transitions(["idle", "stopping", "stopped", "retrying",
-                             "killing", "connecting"])
+        S.valid_transitions(self._BUSY_VALID)
         smgr = self.csf_smgr
         hdl = self.csf_handle
         self.csf_observed = "connected"

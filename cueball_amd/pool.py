@@ -15,6 +15,7 @@ States: starting -> running <-> failed ; stopping -> stopping.backends
 
 from __future__ import annotations
 
+import functools
 import math
 import random
 import time as mod_time
@@ -116,7 +117,7 @@ class _ClaimTicket:
         # stale ('stateChanged' is async): just unlink and skip them;
         # the slot dispatcher copes (lib/pool.js:934-951).
         idleq = pool.p_idleq
-        while len(idleq) > 0:
+        while idleq._len > 0:
             fsm = idleq.shift()
             fsm.p_idleq_node = None
             if not fsm.is_in_state("idle"):
@@ -131,7 +132,7 @@ class _ClaimTicket:
 
         pool.p_waiters.push(handle)
         pool._note_demand()
-        pool._hwm_counter("max-claim-queue", len(pool.p_waiters))
+        pool._hwm_counter("max-claim-queue", pool.p_waiters._len)
         pool._incr_counter("queued-claim")
         pool.rebalance()
 
@@ -308,7 +309,11 @@ class ConnectionPool(FSM):
 
     # -- counters -------------------------------------------------------
     def _incr_counter(self, counter: str) -> None:
-        mod_utils.update_error_metrics(self.p_collector, self.p_uuid, counter)
+        # only tracked error events reach the metrics collector; the
+        # hot counters (claim, queued-claim) skip the call entirely
+        if counter in mod_utils.TRACKED_ERROR_EVENTS:
+            mod_utils.update_error_metrics(self.p_collector, self.p_uuid,
+                                           counter)
         self.p_counters[counter] = self.p_counters.get(counter, 0) + 1
 
     def _hwm_counter(self, counter: str, val: int) -> None:
@@ -319,9 +324,9 @@ class ConnectionPool(FSM):
         """O(1) demand sample on the claim path: fold the current
         busy/extras into the high-water marks consumed by _rebalance
         and _lp_tick (see p_demand_hwm comment in __init__)."""
-        nw = len(self.p_waiters)
-        ni = len(self.p_initq)
-        spares = len(self.p_idleq) + ni - nw
+        nw = self.p_waiters._len
+        ni = self.p_initq._len
+        spares = self.p_idleq._len + ni - nw
         if spares < 0:
             spares = 0
         busy = self.p_total_conns - spares
@@ -721,8 +726,10 @@ class ConnectionPool(FSM):
         fsm.p_initq_node = self.p_initq.push(fsm)
         fsm.p_idleq_node = None
 
-        fsm.on("stateChanged", lambda st: self._slot_state_changed(
-            fsm, key, st))
+        # functools.partial dispatches at C level (the per-event lambda
+        # wrapper showed up in the claim-path profile)
+        fsm.on("stateChanged",
+               functools.partial(self._slot_state_changed, fsm, key))
         fsm.start()
 
     def _slot_state_changed(self, fsm: ConnectionSlotFSM, key: str,
@@ -748,7 +755,7 @@ class ConnectionPool(FSM):
                 return
 
             # Feed waiters, with the CoDel drop check on each.
-            while len(self.p_waiters) > 0:
+            while self.p_waiters._len > 0:
                 hdl = self.p_waiters.shift()
                 drop = self.p_codel is not None and \
                     self.p_codel.overloaded(hdl.ch_started)
