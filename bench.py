@@ -38,25 +38,26 @@ from cueball_amd.resolver import StaticIpResolver  # noqa: E402
 # ---------------------------------------------------------------------------
 # infrastructure
 
+class _EchoProtocol(asyncio.Protocol):
+    """Raw-protocol echo: the asyncio equivalent of node's
+    ``socket.pipe(socket)`` (no streams layer on the hot path)."""
+
+    def connection_made(self, transport):
+        self.transport = transport
+
+    def data_received(self, data):
+        self.transport.write(data)
+
+    def connection_lost(self, exc):
+        pass
+
+
 async def start_backends(n, port0=0):
     """n local TCP echo servers; returns list of (server, port)."""
     servers = []
-
-    async def handle(reader, writer):
-        try:
-            while True:
-                data = await reader.read(65536)
-                if not data:
-                    break
-                writer.write(data)
-                await writer.drain()
-        except (ConnectionResetError, BrokenPipeError):
-            pass
-        finally:
-            writer.close()
-
+    loop = asyncio.get_running_loop()
     for i in range(n):
-        srv = await asyncio.start_server(handle, "127.0.0.1", 0)
+        srv = await loop.create_server(_EchoProtocol, "127.0.0.1", 0)
         port = srv.sockets[0].getsockname()[1]
         servers.append((srv, port))
     return servers
@@ -390,18 +391,52 @@ async def scenario_codel(args, results):
     await asyncio.sleep(0.1)
 
 
+class _FastHttpProtocol(asyncio.Protocol):
+    """GET-only keep-alive HTTP/1.1 backend on a raw protocol (the
+    Python analog of the bare node http.createServer the reference
+    bench uses; the streams-based MockHttpServer stays for tests)."""
+
+    RESP = (b"HTTP/1.1 200 OK\r\nContent-Length: 2\r\n"
+            b"Connection: keep-alive\r\n\r\nok")
+
+    def connection_made(self, transport):
+        self.transport = transport
+        self.buf = b""
+
+    def data_received(self, data):
+        buf = self.buf + data
+        n = buf.count(b"\r\n\r\n")
+        if n:
+            self.transport.write(self.RESP * n)
+            self.buf = buf[buf.rfind(b"\r\n\r\n") + 4:]
+        else:
+            self.buf = buf
+
+    def connection_lost(self, exc):
+        pass
+
+
+class _FastHttpServer:
+    def __init__(self, server, port):
+        self._server = server
+        self.port = port
+
+    def stop(self):
+        self._server.close()
+
+
 async def scenario_agent(args, results):
     """HttpAgent keep-alive over 8 local HTTP backends, 1000 concurrent
     GETs per step (BASELINE config #3)."""
     from cueball_amd.agent import HttpAgent
-    from cueball_amd.testing import DummyResolver, MockHttpServer
+    from cueball_amd.testing import DummyResolver
 
     loop = asyncio.get_running_loop()
     servers = []
     for _ in range(8):
-        s = MockHttpServer()
-        await s.start()
-        servers.append(s)
+        srv = await loop.create_server(_FastHttpProtocol, "127.0.0.1", 0)
+        servers.append(_FastHttpServer(
+            srv, srv.sockets[0].getsockname()[1]))
 
     resolver = DummyResolver()
     agent = HttpAgent({
