@@ -533,27 +533,11 @@ class DNSResolverFSM(FSM):
     # -- AAAA stage -------------------------------------------------------
     @classmethod
     def _get_nics(cls) -> Dict[str, List[Dict[str, Any]]]:
-        """os.networkInterfaces() equivalent: {ifname: [{family,address}]}"""
-        import socket as mod_socket
-        nics: Dict[str, List[Dict[str, Any]]] = {}
-        try:
-            for idx, name in mod_socket.if_nameindex():
-                nics.setdefault(name, [])
-        except OSError:
-            pass
-        try:
-            # getaddrinfo on the hostname gives us configured addresses
-            infos = mod_socket.getaddrinfo(mod_socket.gethostname(), None)
-            for family, _, _, _, sockaddr in infos:
-                if family == mod_socket.AF_INET6:
-                    nics.setdefault("_host", []).append({
-                        "family": "IPv6", "address": sockaddr[0]})
-                elif family == mod_socket.AF_INET:
-                    nics.setdefault("_host", []).append({
-                        "family": "IPv4", "address": sockaddr[0]})
-        except OSError:
-            pass
-        return nics
+        """os.networkInterfaces() equivalent (lib/resolver.js:738-772):
+        real interface enumeration via getifaddrs(3) with
+        /proc/net/if_inet6 and hostname-lookup fallbacks (netif.py)."""
+        from . import netif
+        return netif.network_interfaces()
 
     def state_aaaa(self, S: StateScope) -> None:
         now = self._now_ms()
