@@ -10,6 +10,11 @@ Run: python examples/aiohttp_example.py
 """
 
 import asyncio
+import os
+import sys
+
+sys.path.insert(0, os.path.dirname(os.path.dirname(
+    os.path.abspath(__file__))))
 
 import aiohttp
 
