@@ -71,6 +71,16 @@ __all__ = [
 ]
 
 
+# attach-on-demand debugging (SIGUSR2 / env toggles; the analog of the
+# reference's dtrace-attach detection, lib/utils.js:59-99)
+from . import debug as _debug  # noqa: E402
+
+_debug.apply_env()
+install_attach_handler = _debug.install_attach_handler
+remove_attach_handler = _debug.remove_attach_handler
+__all__ += ["install_attach_handler", "remove_attach_handler"]
+
+
 def __getattr__(name):
     # deferred imports: the agent pulls in the HTTP client machinery
     if name in ("HttpAgent", "HttpsAgent", "PingAgent"):
