@@ -142,13 +142,21 @@ class PoolMonitor:
         obj["resolvers"] = res.r_resolvers
         obj["defaultPort"] = res.r_defport
         obj["state"] = res.get_state()
+        # r_next_* are ms on the (monotonic) event-loop clock; convert
+        # to epoch so the ISO times are real wall-clock expiries like
+        # the reference's Date objects (lib/pool-monitor.js:189-195)
+        loop = getattr(res, "_loop", None)
+        if loop is not None:
+            off = time.time() - loop.time()
+        else:
+            off = 0.0
         nxt: Dict[str, Any] = {}
         if getattr(res, "r_next_service", None):
-            nxt["srv"] = _iso(res.r_next_service)
+            nxt["srv"] = _iso(res.r_next_service / 1000.0 + off)
         if getattr(res, "r_next_v6", None):
-            nxt["v6"] = _iso(res.r_next_v6)
+            nxt["v6"] = _iso(res.r_next_v6 / 1000.0 + off)
         if getattr(res, "r_next_v4", None):
-            nxt["v4"] = _iso(res.r_next_v4)
+            nxt["v4"] = _iso(res.r_next_v4 / 1000.0 + off)
         obj["next"] = nxt
         obj["backends"] = {k: _backend_json(b)
                            for k, b in res.r_backends.items()}
