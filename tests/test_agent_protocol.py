@@ -9,6 +9,18 @@ from cueball_amd.events import EventEmitter
 from conftest import run_vt
 
 
+class Emitter(EventEmitter):
+    """EventEmitter with attribute storage (the pure-Python emitter
+    uses __slots__)."""
+
+    def __init__(self):
+        super().__init__()
+        self.sock = None
+
+    def on_socket(self, s):
+        self.sock = s
+
+
 class FakeHandle:
     def __init__(self):
         self.released = 0
@@ -53,12 +65,10 @@ class FakeAgent:
 def mk(loop):
     agent = FakeAgent(loop)
     pool = FakePool()
-    req = EventEmitter()
-    # HttpRequest surface used by the ticket
-    req.on_socket = lambda s: setattr(req, "sock", s)
+    req = Emitter()
     ticket = _RequestTicket(agent, pool, req)
     hdl = FakeHandle()
-    sock = EventEmitter()
+    sock = Emitter()
     pool.claim_cb(None, hdl, sock)
     return ticket, req, sock, hdl, pool
 
@@ -139,8 +149,7 @@ def test_abort_before_claim_cancels_waiter():
     async def body(loop):
         agent = FakeAgent(loop)
         pool = FakePool()
-        req = EventEmitter()
-        req.on_socket = lambda s: setattr(req, "sock", s)
+        req = Emitter()
         _RequestTicket(agent, pool, req)
 
         req.emit("abort")
