@@ -2564,6 +2564,637 @@ speed_count_listeners(PyObject *mod, PyObject *const *args,
     return PyLong_FromLong(count);
 }
 
+/* ------------------------------------------------------------------ */
+/* Intrusive deque (lib/queue.js): O(1) unlink by node handle.         */
+/* Ownership: the list holds ONE strong ref per linked node; q_next /  */
+/* q_prev / q_queue are borrowed raw pointers, valid exactly while the */
+/* node is linked (q_queue != NULL).                                   */
+
+typedef struct NQueue NQueue;
+
+typedef struct QNode {
+    PyObject_HEAD
+    PyObject *value;
+    struct QNode *q_next;
+    struct QNode *q_prev;
+    NQueue *q_queue;
+} QNode;
+
+struct NQueue {
+    PyObject_HEAD
+    QNode *q_first;     /* borrowed (list ref keeps it alive) */
+    QNode *q_last;      /* borrowed */
+    Py_ssize_t q_len;
+};
+
+extern PyTypeObject QNodeType;
+extern PyTypeObject NQueueType;
+
+static void
+nqueue_unlink(NQueue *q, QNode *n)
+{
+    if (n->q_prev != NULL)
+        n->q_prev->q_next = n->q_next;
+    else
+        q->q_first = n->q_next;
+    if (n->q_next != NULL)
+        n->q_next->q_prev = n->q_prev;
+    else
+        q->q_last = n->q_prev;
+    n->q_next = NULL;
+    n->q_prev = NULL;
+    n->q_queue = NULL;
+    q->q_len--;
+    Py_DECREF((PyObject *)n);   /* drop the list's ref */
+}
+
+static PyObject *
+QNode_remove(PyObject *self_, PyObject *noargs)
+{
+    QNode *n = (QNode *)self_;
+    (void)noargs;
+    if (n->q_queue == NULL) {
+        PyErr_SetString(PyExc_ValueError,
+                        "QueueNode.remove() on unlinked node");
+        return NULL;
+    }
+    nqueue_unlink(n->q_queue, n);
+    Py_RETURN_NONE;
+}
+
+static PyObject *
+QNode_get_linked(PyObject *self_, void *closure)
+{
+    (void)closure;
+    return PyBool_FromLong(((QNode *)self_)->q_queue != NULL);
+}
+
+static PyObject *
+QNode_get_value(PyObject *self_, void *closure)
+{
+    QNode *n = (QNode *)self_;
+    (void)closure;
+    if (n->value == NULL)
+        Py_RETURN_NONE;
+    Py_INCREF(n->value);
+    return n->value;
+}
+
+static int
+QNode_set_value(PyObject *self_, PyObject *v, void *closure)
+{
+    QNode *n = (QNode *)self_;
+    (void)closure;
+    Py_XINCREF(v);
+    Py_XSETREF(n->value, v);
+    return 0;
+}
+
+static int
+QNode_traverse(PyObject *self_, visitproc visit, void *arg)
+{
+    Py_VISIT(((QNode *)self_)->value);
+    return 0;
+}
+
+static int
+QNode_clear_(PyObject *self_)
+{
+    Py_CLEAR(((QNode *)self_)->value);
+    return 0;
+}
+
+static void
+QNode_dealloc(PyObject *self_)
+{
+    PyObject_GC_UnTrack(self_);
+    QNode_clear_(self_);
+    PyObject_GC_Del(self_);
+}
+
+static PyMethodDef QNode_methods[] = {
+    {"remove", QNode_remove, METH_NOARGS, NULL},
+    {NULL, NULL, 0, NULL},
+};
+
+static PyGetSetDef QNode_getset[] = {
+    {"linked", QNode_get_linked, NULL, NULL, NULL},
+    {"value", QNode_get_value, QNode_set_value, NULL, NULL},
+    {NULL, NULL, NULL, NULL, NULL},
+};
+
+PyTypeObject QNodeType = {
+    PyVarObject_HEAD_INIT(NULL, 0)
+    "cueball_amd._speed.QueueNode",
+    sizeof(QNode),
+};
+/* remaining slots filled in queue_types_init() — positional
+ * PyTypeObject initializers are too easy to miscount */
+
+static QNode *
+nqueue_push(NQueue *q, PyObject *value)
+{
+    QNode *n = PyObject_GC_New(QNode, &QNodeType);
+    if (n == NULL)
+        return NULL;
+    Py_INCREF(value);
+    n->value = value;
+    n->q_next = NULL;
+    n->q_prev = q->q_last;
+    n->q_queue = q;
+    PyObject_GC_Track((PyObject *)n);
+    if (q->q_last != NULL)
+        q->q_last->q_next = n;
+    else
+        q->q_first = n;
+    q->q_last = n;
+    q->q_len++;
+    Py_INCREF((PyObject *)n);   /* the list's ref */
+    return n;                   /* caller's ref (from New) */
+}
+
+/* Pop the first value (new ref), or NULL without error if empty. */
+static PyObject *
+nqueue_shift_value(NQueue *q)
+{
+    QNode *n = q->q_first;
+    if (n == NULL)
+        return NULL;
+    PyObject *v = n->value;
+    Py_XINCREF(v);
+    Py_INCREF((PyObject *)n);   /* keep alive across unlink */
+    nqueue_unlink(q, n);
+    Py_DECREF((PyObject *)n);
+    return v ? v : Py_NewRef(Py_None);
+}
+
+static PyObject *
+NQueue_push(PyObject *self_, PyObject *value)
+{
+    return (PyObject *)nqueue_push((NQueue *)self_, value);
+}
+
+static PyObject *
+NQueue_shift(PyObject *self_, PyObject *noargs)
+{
+    (void)noargs;
+    PyObject *v = nqueue_shift_value((NQueue *)self_);
+    if (v == NULL && !PyErr_Occurred())
+        PyErr_SetString(PyExc_IndexError, "shift from empty Queue");
+    return v;
+}
+
+static PyObject *
+NQueue_peek(PyObject *self_, PyObject *noargs)
+{
+    NQueue *q = (NQueue *)self_;
+    (void)noargs;
+    if (q->q_first == NULL) {
+        PyErr_SetString(PyExc_IndexError, "peek from empty Queue");
+        return NULL;
+    }
+    PyObject *v = q->q_first->value;
+    if (v == NULL)
+        Py_RETURN_NONE;
+    Py_INCREF(v);
+    return v;
+}
+
+static PyObject *
+NQueue_is_empty(PyObject *self_, PyObject *noargs)
+{
+    (void)noargs;
+    return PyBool_FromLong(((NQueue *)self_)->q_len == 0);
+}
+
+static PyObject *
+NQueue_for_each(PyObject *self_, PyObject *cb)
+{
+    NQueue *q = (NQueue *)self_;
+    QNode *n = q->q_first;
+    Py_XINCREF((PyObject *)n);
+    while (n != NULL) {
+        QNode *nxt = n->q_next;
+        Py_XINCREF((PyObject *)nxt);
+        PyObject *args[2] = {n->value ? n->value : Py_None,
+                             (PyObject *)n};
+        PyObject *r = PyObject_Vectorcall(cb, args, 2, NULL);
+        Py_DECREF((PyObject *)n);
+        if (r == NULL) {
+            Py_XDECREF((PyObject *)nxt);
+            return NULL;
+        }
+        Py_DECREF(r);
+        /* if the callback unlinked the captured successor, stop (it
+         * no longer belongs to this list) — matches the pure twin's
+         * generator behavior */
+        if (nxt != NULL && nxt->q_queue != q) {
+            Py_DECREF((PyObject *)nxt);
+            break;
+        }
+        n = nxt;
+    }
+    Py_RETURN_NONE;
+}
+
+static PyObject *
+NQueue_iter(PyObject *self_)
+{
+    /* snapshot the values; O(1)-removal users iterate rarely */
+    NQueue *q = (NQueue *)self_;
+    PyObject *list = PyList_New(0);
+    if (list == NULL)
+        return NULL;
+    for (QNode *n = q->q_first; n != NULL; n = n->q_next) {
+        if (PyList_Append(list, n->value ? n->value : Py_None) < 0) {
+            Py_DECREF(list);
+            return NULL;
+        }
+    }
+    PyObject *it = PyObject_GetIter(list);
+    Py_DECREF(list);
+    return it;
+}
+
+static Py_ssize_t
+NQueue_len(PyObject *self_)
+{
+    return ((NQueue *)self_)->q_len;
+}
+
+static PyObject *
+NQueue_get_len(PyObject *self_, void *closure)
+{
+    (void)closure;
+    return PyLong_FromSsize_t(((NQueue *)self_)->q_len);
+}
+
+static int
+NQueue_traverse(PyObject *self_, visitproc visit, void *arg)
+{
+    NQueue *q = (NQueue *)self_;
+    for (QNode *n = q->q_first; n != NULL; n = n->q_next)
+        Py_VISIT((PyObject *)n);
+    return 0;
+}
+
+static int
+NQueue_clear_(PyObject *self_)
+{
+    NQueue *q = (NQueue *)self_;
+    while (q->q_first != NULL) {
+        QNode *n = q->q_first;
+        Py_INCREF((PyObject *)n);
+        nqueue_unlink(q, n);
+        Py_DECREF((PyObject *)n);
+    }
+    return 0;
+}
+
+static void
+NQueue_dealloc(PyObject *self_)
+{
+    PyObject_GC_UnTrack(self_);
+    NQueue_clear_(self_);
+    PyObject_GC_Del(self_);
+}
+
+static PyObject *
+NQueue_new(PyTypeObject *type, PyObject *args, PyObject *kwds)
+{
+    (void)args; (void)kwds;
+    NQueue *q = PyObject_GC_New(NQueue, type);
+    if (q == NULL)
+        return NULL;
+    q->q_first = NULL;
+    q->q_last = NULL;
+    q->q_len = 0;
+    PyObject_GC_Track((PyObject *)q);
+    return (PyObject *)q;
+}
+
+static PyMethodDef NQueue_methods[] = {
+    {"push", NQueue_push, METH_O, NULL},
+    {"shift", NQueue_shift, METH_NOARGS, NULL},
+    {"peek", NQueue_peek, METH_NOARGS, NULL},
+    {"is_empty", NQueue_is_empty, METH_NOARGS, NULL},
+    {"for_each", NQueue_for_each, METH_O, NULL},
+    {NULL, NULL, 0, NULL},
+};
+
+static PyGetSetDef NQueue_getset[] = {
+    {"_len", NQueue_get_len, NULL, NULL, NULL},
+    {"length", NQueue_get_len, NULL, NULL, NULL},
+    {NULL, NULL, NULL, NULL, NULL},
+};
+
+static PySequenceMethods NQueue_as_sequence = {
+    NQueue_len,                 /* sq_length */
+};
+
+PyTypeObject NQueueType = {
+    PyVarObject_HEAD_INIT(NULL, 0)
+    "cueball_amd._speed.Queue",
+    sizeof(NQueue),
+};
+
+static void
+queue_types_init(void)
+{
+    QNodeType.tp_dealloc = QNode_dealloc;
+    QNodeType.tp_flags = Py_TPFLAGS_DEFAULT | Py_TPFLAGS_HAVE_GC;
+    QNodeType.tp_traverse = QNode_traverse;
+    QNodeType.tp_clear = QNode_clear_;
+    QNodeType.tp_methods = QNode_methods;
+    QNodeType.tp_getset = QNode_getset;
+
+    NQueueType.tp_dealloc = NQueue_dealloc;
+    NQueueType.tp_as_sequence = &NQueue_as_sequence;
+    NQueueType.tp_flags = Py_TPFLAGS_DEFAULT | Py_TPFLAGS_HAVE_GC;
+    NQueueType.tp_traverse = NQueue_traverse;
+    NQueueType.tp_clear = NQueue_clear_;
+    NQueueType.tp_iter = NQueue_iter;
+    NQueueType.tp_methods = NQueue_methods;
+    NQueueType.tp_getset = NQueue_getset;
+    NQueueType.tp_new = NQueue_new;
+}
+
+/* ------------------------------------------------------------------ */
+/* ClaimTicket: the pool's per-claim driver (pool.py _ClaimTicket) in  */
+/* C.  Registered as the claim handle's stateChanged listener; on      */
+/* every return to "waiting" it scans the idle queue and performs the  */
+/* try_ handoff without entering Python.  Anything off the hot path    */
+/* (pool not running, idle queue empty, errorOnEmpty) delegates to     */
+/* pool._ticket_slow, which holds the full reference logic.            */
+
+PyObject *s_running_st;      /* "running" */
+PyObject *s_p_idleq_node;    /* "p_idleq_node" */
+PyObject *s_p_total_conns;   /* "p_total_conns" */
+PyObject *s_p_busy_hwm;      /* "p_busy_hwm" */
+PyObject *s_p_demand_hwm;    /* "p_demand_hwm" */
+PyObject *s_ticket_slow;     /* "_ticket_slow" */
+
+typedef struct {
+    PyObject_HEAD
+    PyObject *ct_pool;
+    PyObject *ct_handle;     /* a CHOb */
+    NQueue *ct_idleq;        /* owned; NULL => always delegate */
+    NQueue *ct_waiters;      /* owned */
+    NQueue *ct_initq;        /* owned */
+    int ct_err_on_empty;
+} CTOb;
+
+extern PyTypeObject CTType;
+
+static int
+ct_note_demand(CTOb *t)
+{
+    /* O(1) inline of pool._note_demand() */
+    Py_ssize_t nw = t->ct_waiters->q_len;
+    Py_ssize_t ni = t->ct_initq->q_len;
+    Py_ssize_t spares = t->ct_idleq->q_len + ni - nw;
+    if (spares < 0)
+        spares = 0;
+    PyObject *tot = PyObject_GetAttr(t->ct_pool, s_p_total_conns);
+    if (tot == NULL)
+        return -1;
+    long total = PyLong_AsLong(tot);
+    Py_DECREF(tot);
+    if (total == -1 && PyErr_Occurred())
+        return -1;
+    long busy = total - (long)spares;
+    if (busy < 0)
+        busy = 0;
+    long extras = (long)(nw - ni);
+    if (extras < 0)
+        extras = 0;
+
+    PyObject *cur = PyObject_GetAttr(t->ct_pool, s_p_busy_hwm);
+    if (cur == NULL)
+        return -1;
+    long curv = PyLong_AsLong(cur);
+    Py_DECREF(cur);
+    if (curv == -1 && PyErr_Occurred())
+        return -1;
+    if (busy > curv) {
+        PyObject *nv = PyLong_FromLong(busy);
+        if (nv == NULL)
+            return -1;
+        int r = PyObject_SetAttr(t->ct_pool, s_p_busy_hwm, nv);
+        Py_DECREF(nv);
+        if (r < 0)
+            return -1;
+    }
+    cur = PyObject_GetAttr(t->ct_pool, s_p_demand_hwm);
+    if (cur == NULL)
+        return -1;
+    curv = PyLong_AsLong(cur);
+    Py_DECREF(cur);
+    if (curv == -1 && PyErr_Occurred())
+        return -1;
+    if (busy + extras > curv) {
+        PyObject *nv = PyLong_FromLong(busy + extras);
+        if (nv == NULL)
+            return -1;
+        int r = PyObject_SetAttr(t->ct_pool, s_p_demand_hwm, nv);
+        Py_DECREF(nv);
+        if (r < 0)
+            return -1;
+    }
+    return 0;
+}
+
+static int
+ct_slow(CTOb *t)
+{
+    PyObject *r = PyObject_CallMethodObjArgs(
+        t->ct_pool, s_ticket_slow, t->ct_handle,
+        t->ct_err_on_empty ? Py_True : Py_False, NULL);
+    if (r == NULL)
+        return -1;
+    Py_DECREF(r);
+    return 0;
+}
+
+static int
+ct_try_next(CTOb *t)
+{
+    CHOb *h = (CHOb *)t->ct_handle;
+    if (!ch_state_is(h, s_waiting))
+        return 0;
+    if (t->ct_idleq == NULL)
+        return ct_slow(t);
+    PyObject *pst = ((FSMOb *)t->ct_pool)->f_state;
+    if (!(pst == s_running_st ||
+          (pst != NULL && PyUnicode_Compare(pst, s_running_st) == 0)))
+        return ct_slow(t);
+
+    NQueue *q = t->ct_idleq;
+    while (q->q_len > 0) {
+        QNode *n = q->q_first;
+        PyObject *fsm = n->value;
+        Py_XINCREF(fsm);
+        Py_INCREF((PyObject *)n);
+        nqueue_unlink(q, n);
+        Py_DECREF((PyObject *)n);
+        if (fsm == NULL)
+            continue;
+        if (PyObject_SetAttr(fsm, s_p_idleq_node, Py_None) < 0) {
+            Py_DECREF(fsm);
+            return -1;
+        }
+        /* stale entries tolerated: only a slot still in 'idle' is
+         * usable (lib/pool.js:934-951) */
+        int isidle;
+        if (PyObject_TypeCheck(fsm, &FSMType)) {
+            PyObject *fs = ((FSMOb *)fsm)->f_state;
+            isidle = (fs == s_idle ||
+                      (fs != NULL &&
+                       PyUnicode_Compare(fs, s_idle) == 0));
+        } else {
+            PyObject *r = PyObject_CallMethodObjArgs(
+                fsm, s_is_in_state, s_idle, NULL);
+            if (r == NULL) {
+                Py_DECREF(fsm);
+                return -1;
+            }
+            isidle = PyObject_IsTrue(r);
+            Py_DECREF(r);
+        }
+        if (!isidle) {
+            Py_DECREF(fsm);
+            continue;
+        }
+        PyObject *r = CH_try_((PyObject *)h, fsm);
+        Py_DECREF(fsm);
+        if (r == NULL)
+            return -1;
+        Py_DECREF(r);
+        return ct_note_demand(t);
+    }
+    return ct_slow(t);
+}
+
+static PyObject *
+CT_call(PyObject *self_, PyObject *args, PyObject *kwds)
+{
+    CTOb *t = (CTOb *)self_;
+    (void)kwds;
+    PyObject *st;
+    if (PyTuple_GET_SIZE(args) != 1) {
+        PyErr_SetString(PyExc_TypeError, "ClaimTicket(state)");
+        return NULL;
+    }
+    st = PyTuple_GET_ITEM(args, 0);
+    if (st == s_waiting ||
+        (PyUnicode_Check(st) && PyUnicode_Compare(st, s_waiting) == 0)) {
+        if (ct_try_next(t) < 0)
+            return NULL;
+    }
+    Py_RETURN_NONE;
+}
+
+static int
+CT_traverse(PyObject *self_, visitproc visit, void *arg)
+{
+    CTOb *t = (CTOb *)self_;
+    Py_VISIT(t->ct_pool);
+    Py_VISIT(t->ct_handle);
+    Py_VISIT((PyObject *)t->ct_idleq);
+    Py_VISIT((PyObject *)t->ct_waiters);
+    Py_VISIT((PyObject *)t->ct_initq);
+    return 0;
+}
+
+static int
+CT_clear_(PyObject *self_)
+{
+    CTOb *t = (CTOb *)self_;
+    Py_CLEAR(t->ct_pool);
+    Py_CLEAR(t->ct_handle);
+    Py_CLEAR(t->ct_idleq);
+    Py_CLEAR(t->ct_waiters);
+    Py_CLEAR(t->ct_initq);
+    return 0;
+}
+
+static void
+CT_dealloc(PyObject *self_)
+{
+    PyObject_GC_UnTrack(self_);
+    CT_clear_(self_);
+    PyObject_GC_Del(self_);
+}
+
+static PyObject *
+CT_new(PyTypeObject *type, PyObject *args, PyObject *kwds)
+{
+    PyObject *pool, *handle, *err_on_empty;
+    (void)kwds;
+    if (!PyArg_ParseTuple(args, "OOO", &pool, &handle, &err_on_empty))
+        return NULL;
+    if (!PyObject_TypeCheck(pool, &FSMType)) {
+        PyErr_SetString(PyExc_TypeError, "pool must be a native FSM");
+        return NULL;
+    }
+    if (!PyObject_TypeCheck(handle, &CHType)) {
+        PyErr_SetString(PyExc_TypeError,
+                        "handle must be a ClaimHandleBase");
+        return NULL;
+    }
+    CTOb *t = PyObject_GC_New(CTOb, type);
+    if (t == NULL)
+        return NULL;
+    Py_INCREF(pool);
+    t->ct_pool = pool;
+    Py_INCREF(handle);
+    t->ct_handle = handle;
+    t->ct_idleq = NULL;
+    t->ct_waiters = NULL;
+    t->ct_initq = NULL;
+    t->ct_err_on_empty = PyObject_IsTrue(err_on_empty);
+    PyObject_GC_Track((PyObject *)t);
+
+    /* cache the pool's queues; if they are not the native deque the
+     * ticket permanently delegates to the python slow path */
+    PyObject *iq = PyObject_GetAttrString(pool, "p_idleq");
+    PyObject *wq = iq ? PyObject_GetAttrString(pool, "p_waiters") : NULL;
+    PyObject *nq = wq ? PyObject_GetAttrString(pool, "p_initq") : NULL;
+    if (nq == NULL) {
+        PyErr_Clear();
+        Py_XDECREF(iq);
+        Py_XDECREF(wq);
+    } else if (PyObject_TypeCheck(iq, &NQueueType) &&
+               PyObject_TypeCheck(wq, &NQueueType) &&
+               PyObject_TypeCheck(nq, &NQueueType)) {
+        t->ct_idleq = (NQueue *)iq;
+        t->ct_waiters = (NQueue *)wq;
+        t->ct_initq = (NQueue *)nq;
+    } else {
+        Py_DECREF(iq);
+        Py_DECREF(wq);
+        Py_DECREF(nq);
+    }
+    return (PyObject *)t;
+}
+
+PyTypeObject CTType = {
+    PyVarObject_HEAD_INIT(NULL, 0)
+    "cueball_amd._speed.ClaimTicket",
+    sizeof(CTOb),
+};
+
+static void
+ct_type_init(void)
+{
+    CTType.tp_dealloc = CT_dealloc;
+    CTType.tp_call = CT_call;
+    CTType.tp_flags = Py_TPFLAGS_DEFAULT | Py_TPFLAGS_HAVE_GC;
+    CTType.tp_traverse = CT_traverse;
+    CTType.tp_clear = CT_clear_;
+    CTType.tp_new = CT_new;
+}
+
 PyObject *
 speed_set_tracer(PyObject *mod, PyObject *fn)
 {
@@ -2646,6 +3277,14 @@ PyInit__speed(void)
     if (g_flush_batches == NULL)
         return NULL;
 
+    s_running_st = PyUnicode_InternFromString("running");
+    s_p_idleq_node = PyUnicode_InternFromString("p_idleq_node");
+    s_p_total_conns = PyUnicode_InternFromString("p_total_conns");
+    s_p_busy_hwm = PyUnicode_InternFromString("p_busy_hwm");
+    s_p_demand_hwm = PyUnicode_InternFromString("p_demand_hwm");
+    s_ticket_slow = PyUnicode_InternFromString("_ticket_slow");
+    queue_types_init();
+    ct_type_init();
     if (PyType_Ready(&EmitterType) < 0 ||
         PyType_Ready(&OnceWrapperType) < 0 ||
         PyType_Ready(&GuardedCbType) < 0 ||
@@ -2655,7 +3294,10 @@ PyInit__speed(void)
         PyType_Ready(&ConnErrCbType) < 0 ||
         PyType_Ready(&FailCbType) < 0 ||
         PyType_Ready(&CHType) < 0 ||
-        PyType_Ready(&FSMType) < 0)
+        PyType_Ready(&FSMType) < 0 ||
+        PyType_Ready(&QNodeType) < 0 ||
+        PyType_Ready(&NQueueType) < 0 ||
+        PyType_Ready(&CTType) < 0)
         return NULL;
 
     g_remove_desc = PyDict_GetItemString(EmitterType.tp_dict,
@@ -2675,5 +3317,11 @@ PyInit__speed(void)
     PyModule_AddObject(m, "FSM", (PyObject *)&FSMType);
     Py_INCREF(&CHType);
     PyModule_AddObject(m, "ClaimHandleBase", (PyObject *)&CHType);
+    Py_INCREF(&NQueueType);
+    PyModule_AddObject(m, "Queue", (PyObject *)&NQueueType);
+    Py_INCREF(&QNodeType);
+    PyModule_AddObject(m, "QueueNode", (PyObject *)&QNodeType);
+    Py_INCREF(&CTType);
+    PyModule_AddObject(m, "ClaimTicket", (PyObject *)&CTType);
     return m;
 }

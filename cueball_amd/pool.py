@@ -17,6 +17,7 @@ from __future__ import annotations
 
 import functools
 import math
+import os as _os
 import random
 import time as mod_time
 import uuid as mod_uuid
@@ -81,9 +82,14 @@ class FIRFilter:
 
 
 class _ClaimTicket:
-    """Per-claim driver: retries try_next whenever the handle returns to
-    'waiting' (registered as the handle's stateChanged listener; the
-    closure-free form of lib/pool.js:922-968, on the claim hot path)."""
+    """Per-claim driver: retries the claim whenever the handle returns
+    to 'waiting' (registered as the handle's stateChanged listener; the
+    closure-free form of lib/pool.js:922-968, on the claim hot path).
+
+    When the native core is loaded, ``_speed.ClaimTicket`` replaces
+    this class on the claim path: it performs the idle-queue handoff
+    entirely in C and delegates everything else to the same
+    ``pool._ticket_slow`` this class uses."""
 
     __slots__ = ("pool", "handle", "err_on_empty")
 
@@ -95,46 +101,15 @@ class _ClaimTicket:
 
     def __call__(self, st: str) -> None:
         if st == "waiting":
-            self.try_next()
+            self.pool._ticket_slow(self.handle, self.err_on_empty)
 
-    def try_next(self) -> None:
-        pool = self.pool
-        handle = self.handle
-        if not handle.is_in_state("waiting"):
-            return
-        # The first try runs on the next loop turn; the pool may have
-        # started stopping (or failed) in between — fail now rather
-        # than queueing a waiter nothing will ever feed (companion to
-        # the stopping-state waiter drain).
-        if pool.is_in_state("stopping") or pool.is_in_state("stopped"):
-            handle.fail(mod_errors.PoolStoppingError(pool))
-            return
-        if pool.is_in_state("failed"):
-            handle.fail(mod_errors.PoolFailedError(
-                pool, pool.p_last_error))
-            return
-        # Idle connections sitting around?  Take one.  Entries may be
-        # stale ('stateChanged' is async): just unlink and skip them;
-        # the slot dispatcher copes (lib/pool.js:934-951).
-        idleq = pool.p_idleq
-        while idleq._len > 0:
-            fsm = idleq.shift()
-            fsm.p_idleq_node = None
-            if not fsm.is_in_state("idle"):
-                continue
-            handle.try_(fsm)
-            pool._note_demand()
-            return
 
-        if self.err_on_empty and pool.p_resolver.count() < 1:
-            handle.fail(mod_errors.NoBackendsError(
-                pool, pool.p_resolver.get_last_error()))
-
-        pool.p_waiters.push(handle)
-        pool._note_demand()
-        pool._hwm_counter("max-claim-queue", pool.p_waiters._len)
-        pool._incr_counter("queued-claim")
-        pool.rebalance()
+_NativeClaimTicket = None
+if not _os.environ.get("CUEBALL_PURE"):
+    try:
+        from ._speed import ClaimTicket as _NativeClaimTicket  # noqa: F811
+    except ImportError:
+        pass
 
 
 class _CancelStub:
@@ -831,6 +806,48 @@ class ConnectionPool(FSM):
             "waiterCount": len(self.p_waiters),
         }
 
+    def _ticket_slow(self, handle: ClaimHandle,
+                     err_on_empty: bool) -> None:
+        """The full claim-retry logic (lib/pool.js:922-968): invoked on
+        every handle return to 'waiting' — directly in pure mode, and
+        as the native ClaimTicket's off-hot-path fallback (pool not in
+        'running', or the idle queue empty)."""
+        if not handle.is_in_state("waiting"):
+            return
+        # The first try runs on the next loop turn; the pool may have
+        # started stopping (or failed) in between — fail now rather
+        # than queueing a waiter nothing will ever feed (companion to
+        # the stopping-state waiter drain).
+        if self.is_in_state("stopping") or self.is_in_state("stopped"):
+            handle.fail(mod_errors.PoolStoppingError(self))
+            return
+        if self.is_in_state("failed"):
+            handle.fail(mod_errors.PoolFailedError(
+                self, self.p_last_error))
+            return
+        # Idle connections sitting around?  Take one.  Entries may be
+        # stale ('stateChanged' is async): just unlink and skip them;
+        # the slot dispatcher copes (lib/pool.js:934-951).
+        idleq = self.p_idleq
+        while idleq._len > 0:
+            fsm = idleq.shift()
+            fsm.p_idleq_node = None
+            if not fsm.is_in_state("idle"):
+                continue
+            handle.try_(fsm)
+            self._note_demand()
+            return
+
+        if err_on_empty and self.p_resolver.count() < 1:
+            handle.fail(mod_errors.NoBackendsError(
+                self, self.p_resolver.get_last_error()))
+
+        self.p_waiters.push(handle)
+        self._note_demand()
+        self._hwm_counter("max-claim-queue", self.p_waiters._len)
+        self._incr_counter("queued-claim")
+        self.rebalance()
+
     def claim(self, options: Any = None, cb: Optional[Callable] = None):
         """Claim a connection: cb(err, handle, connection).
 
@@ -871,7 +888,11 @@ class ConnectionPool(FSM):
         # stateChanged('waiting'); the ticket receives it on the next
         # loop turn and runs the first try_next then — claim() never
         # fires the callback synchronously (lib/pool.js:922-968).
-        handle.on("stateChanged", _ClaimTicket(self, handle, err_on_empty))
+        if _NativeClaimTicket is not None:
+            ticket = _NativeClaimTicket(self, handle, err_on_empty)
+        else:
+            ticket = _ClaimTicket(self, handle, err_on_empty)
+        handle.on("stateChanged", ticket)
         return handle
 
     def _claim_shortcircuit(self, cb: Callable, err: BaseException):

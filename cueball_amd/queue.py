@@ -97,3 +97,19 @@ class Queue:
     @property
     def length(self) -> int:
         return self._len
+
+
+# ---------------------------------------------------------------------------
+# Native core: when the C++ extension is built, its intrusive deque
+# replaces the pure-Python one above — identical semantics (the suite
+# runs against either; CUEBALL_PURE=1 forces the Python classes).
+import os as _os
+
+PurePythonQueue = Queue
+PurePythonQueueNode = QueueNode
+
+if not _os.environ.get("CUEBALL_PURE"):
+    try:
+        from ._speed import Queue, QueueNode  # type: ignore # noqa: F811
+    except ImportError:
+        pass
