@@ -582,6 +582,31 @@ scope_check_active(Scope *self, const char *what)
     return 0;
 }
 
+/* the registration core of S.on(), callable from C (SlotKit) */
+int
+scope_on_c(Scope *self, PyObject *emitter, PyObject *event, PyObject *cb)
+{
+    PyObject *r;
+    if (Py_TYPE(emitter) == &EmitterType) {
+        r = emitter_add((Emitter *)emitter, event, cb);
+    } else {
+        r = PyObject_CallMethodObjArgs(emitter, s_on, event, cb, NULL);
+    }
+    if (r == NULL)
+        return -1;
+    Py_DECREF(r);
+    if (self->sc_listeners == NULL) {
+        self->sc_listeners = PyList_New(0);
+        if (self->sc_listeners == NULL)
+            return -1;
+    }
+    if (PyList_Append(self->sc_listeners, emitter) < 0 ||
+        PyList_Append(self->sc_listeners, event) < 0 ||
+        PyList_Append(self->sc_listeners, cb) < 0)
+        return -1;
+    return 0;
+}
+
 PyObject *
 Scope_on(PyObject *self_, PyObject *const *args, Py_ssize_t nargs)
 {
@@ -852,9 +877,17 @@ struct FSMOb {
     PyObject *f_emit_queue;     /* list */
     PyObject *f_history;        /* list */
     PyObject *f_flush_bound;    /* cached bound _flush_state_changed */
+    PyObject *f_fastkit;        /* SlotKit* or NULL: native entries for
+                                 * the busy/idle hot cycle */
     int f_entering;
     int f_emit_scheduled;
 };
+
+/* forward decl (SlotKit lives after the CH/Queue sections);
+ * returns 1 = state entered natively, 0 = use the python entry,
+ * -1 = error */
+int slotkit_maybe_entry(PyObject *kit_, FSMOb *fsm, PyObject *target,
+                        Scope *scope);
 
 extern PyTypeObject FSMType;
 
@@ -1161,29 +1194,48 @@ fsm_enter_loop(FSMOb *self, PyObject *state)
             Py_DECREF(tr);
         }
 
-        PyObject *attr = entry_attr_name(target);
-        if (attr == NULL) {
-            Py_DECREF(scope);
-            Py_DECREF(target);
-            return -1;
+        int handled = 0;
+        if (self->f_fastkit != NULL) {
+            self->f_entering = 1;
+            handled = slotkit_maybe_entry(self->f_fastkit, self, target,
+                                          scope);
+            self->f_entering = 0;
+            if (handled < 0) {
+                Py_DECREF(scope);
+                Py_DECREF(target);
+                return -1;
+            }
         }
-        PyObject *entry = PyObject_GetAttr((PyObject *)self, attr);
-        Py_DECREF(attr);
-        if (entry == NULL) {
-            PyErr_Clear();
-            PyErr_Format(g_fsm_error,
-                         "%s has no state-entry function for %R",
-                         Py_TYPE(self)->tp_name, target);
+        PyObject *r;
+        if (handled) {
             Py_DECREF(scope);
-            Py_DECREF(target);
-            return -1;
-        }
+            r = Py_None;
+            Py_INCREF(r);
+        } else {
+            PyObject *attr = entry_attr_name(target);
+            if (attr == NULL) {
+                Py_DECREF(scope);
+                Py_DECREF(target);
+                return -1;
+            }
+            PyObject *entry = PyObject_GetAttr((PyObject *)self, attr);
+            Py_DECREF(attr);
+            if (entry == NULL) {
+                PyErr_Clear();
+                PyErr_Format(g_fsm_error,
+                             "%s has no state-entry function for %R",
+                             Py_TYPE(self)->tp_name, target);
+                Py_DECREF(scope);
+                Py_DECREF(target);
+                return -1;
+            }
 
-        self->f_entering = 1;
-        PyObject *r = PyObject_CallOneArg(entry, (PyObject *)scope);
-        Py_DECREF(entry);
-        Py_DECREF(scope);
-        self->f_entering = 0;
+            self->f_entering = 1;
+            r = PyObject_CallOneArg(entry, (PyObject *)scope);
+            Py_DECREF(entry);
+            Py_DECREF(scope);
+            self->f_entering = 0;
+        }
         PyObject *pend = self->f_pending;
         self->f_pending = NULL;
         if (r == NULL) {
@@ -1531,6 +1583,7 @@ FSM_traverse(PyObject *self_, visitproc visit, void *arg)
     Py_VISIT(self->f_emit_queue);
     Py_VISIT(self->f_history);
     Py_VISIT(self->f_flush_bound);
+    Py_VISIT(self->f_fastkit);
     return Emitter_traverse(self_, visit, arg);
 }
 
@@ -1546,6 +1599,7 @@ FSM_clear_(PyObject *self_)
     Py_CLEAR(self->f_emit_queue);
     Py_CLEAR(self->f_history);
     Py_CLEAR(self->f_flush_bound);
+    Py_CLEAR(self->f_fastkit);
     return Emitter_clear_(self_);
 }
 
@@ -1561,12 +1615,26 @@ FSM_dealloc(PyObject *self_)
     tp->tp_free(self_);
 }
 
+static PyObject *
+FSM_set_fast_kit(PyObject *self_, PyObject *kit)
+{
+    FSMOb *self = (FSMOb *)self_;
+    if (kit == Py_None) {
+        Py_CLEAR(self->f_fastkit);
+    } else {
+        Py_INCREF(kit);
+        Py_XSETREF(self->f_fastkit, kit);
+    }
+    Py_RETURN_NONE;
+}
+
 PyMethodDef FSM_methods[] = {
     {"get_state", FSM_get_state, METH_NOARGS, NULL},
     {"is_in_state", FSM_is_in_state, METH_O, NULL},
     {"get_state_history", FSM_get_state_history, METH_NOARGS, NULL},
     {"goto_state", FSM_goto_state_py, METH_O, NULL},
     {"_flush_state_changed", FSM_flush_state_changed, METH_NOARGS, NULL},
+    {"_set_fast_kit", FSM_set_fast_kit, METH_O, NULL},
     {NULL, NULL, 0, NULL},
 };
 
@@ -3195,6 +3263,431 @@ ct_type_init(void)
     CTType.tp_new = CT_new;
 }
 
+/* ------------------------------------------------------------------ */
+/* SlotKit: native entries for ConnectionSlotFSM's busy/idle hot       */
+/* cycle (connection_fsm.py state_busy/state_idle).  One claim/release */
+/* runs idle -> busy -> idle; with the kit installed both entries and  */
+/* their event callbacks execute in C, with zero per-cycle Python      */
+/* allocation.  Installed only for slots without a health checker;     */
+/* monitor/unwanted idle entries fall back to the Python entry, and    */
+/* all exit transitions out of the cycle enter Python states.          */
+
+PyObject *s_connected_st;    /* "connected" */
+PyObject *s_busy_st;         /* "busy" */
+PyObject *s_stopping_st;     /* "stopping" */
+PyObject *s_stopped_st;      /* "stopped" */
+PyObject *s_retrying_st;     /* "retrying" */
+PyObject *s_connecting_st;   /* "connecting" */
+PyObject *s_killing_st;      /* "killing" */
+PyObject *s_unwanted_evt;    /* "unwanted" */
+PyObject *s_csf_handle;      /* "csf_handle" */
+PyObject *s_csf_prev_handle; /* "csf_prev_handle" */
+PyObject *s_csf_wanted;      /* "csf_wanted" */
+PyObject *s_csf_monitor;     /* "csf_monitor" */
+PyObject *s_sm_socket;       /* "sm_socket" */
+PyObject *t_busy_valid;      /* like _BUSY_VALID */
+PyObject *t_idle_valid;      /* like _IDLE_VALID */
+
+typedef struct SlotKit SlotKit;
+
+/* preallocated listener callables; `which` selects the behavior */
+enum { KCB_BUSY_SMGR = 0, KCB_BUSY_HDL, KCB_IDLE_SMGR, KCB_IDLE_UNWANTED };
+
+typedef struct {
+    PyObject_HEAD
+    SlotKit *kc_kit;            /* owned */
+    int kc_which;
+} KitCb;
+
+struct SlotKit {
+    PyObject_HEAD
+    FSMOb *k_slot;              /* owned */
+    PyObject *k_smgr;           /* owned; an FSMOb */
+    PyObject *k_observed;       /* event-observed smgr state, owned */
+    Scope *k_busy_scope;        /* owned; scope of the last busy entry */
+    Scope *k_idle_scope;        /* owned; scope of the last idle entry */
+    PyObject *k_cbs[4];         /* owned KitCb instances */
+};
+
+extern PyTypeObject SlotKitType;
+extern PyTypeObject KitCbType;
+
+static int
+kit_truthy_attr(PyObject *obj, PyObject *name)
+{
+    PyObject *v = PyObject_GetAttr(obj, name);
+    if (v == NULL)
+        return -1;
+    int t = PyObject_IsTrue(v);
+    Py_DECREF(v);
+    return t;
+}
+
+static int
+kit_scope_goto(Scope *scope, PyObject *state)
+{
+    if (scope == NULL || !scope->sc_active)
+        return 0;  /* stale handler in the same cascade: obsolete */
+    return fsm_goto_state((FSMOb *)scope->sc_fsm, state);
+}
+
+static int
+kit_busy_on_release(SlotKit *k)
+{
+    Scope *S = k->k_busy_scope;
+    PyObject *st = k->k_observed;
+    int wanted = kit_truthy_attr((PyObject *)k->k_slot, s_csf_wanted);
+    if (wanted < 0)
+        return -1;
+    if (st == s_connected_st ||
+        (st != NULL && PyUnicode_Compare(st, s_connected_st) == 0))
+        return kit_scope_goto(S, wanted ? s_idle : s_stopping_st);
+    if (st == s_closed || PyUnicode_Compare(st, s_closed) == 0)
+        return kit_scope_goto(S, wanted ? s_connecting_st : s_stopped_st);
+    if (st == s_error_evt || PyUnicode_Compare(st, s_error_evt) == 0)
+        return kit_scope_goto(S, s_retrying_st);
+    PyErr_Format(g_fsm_error,
+                 "Handle released while smgr was in unhandled state %R",
+                 st ? st : Py_None);
+    return -1;
+}
+
+static int
+kit_busy_on_close(SlotKit *k)
+{
+    Scope *S = k->k_busy_scope;
+    PyObject *st = k->k_observed;
+    if (st == s_connected_st ||
+        (st != NULL && PyUnicode_Compare(st, s_connected_st) == 0))
+        return kit_scope_goto(S, s_killing_st);
+    return kit_scope_goto(S, s_retrying_st);
+}
+
+static int
+kit_idle_smgr_changed(SlotKit *k, PyObject *st)
+{
+    Scope *S = k->k_idle_scope;
+    int wanted = kit_truthy_attr((PyObject *)k->k_slot, s_csf_wanted);
+    if (wanted < 0)
+        return -1;
+    if (st == s_error_evt || PyUnicode_Compare(st, s_error_evt) == 0)
+        return kit_scope_goto(S, wanted ? s_retrying_st : s_stopped_st);
+    if (st == s_closed || PyUnicode_Compare(st, s_closed) == 0)
+        return kit_scope_goto(S, wanted ? s_connecting_st : s_stopped_st);
+    PyErr_Format(g_fsm_error,
+                 "Unhandled smgr state transition: connected => %R", st);
+    return -1;
+}
+
+static int
+kit_idle_unwanted(SlotKit *k)
+{
+    /* mirror of _idle_on_unwanted incl. the same-spin death fix */
+    Scope *S = k->k_idle_scope;
+    PyObject *sst = ((FSMOb *)k->k_smgr)->f_state;
+    if (sst == s_connected_st ||
+        (sst != NULL && PyUnicode_Compare(sst, s_connected_st) == 0))
+        return kit_scope_goto(S, s_stopping_st);
+    if (sst == s_error_evt || sst == s_closed ||
+        (sst != NULL && (PyUnicode_Compare(sst, s_error_evt) == 0 ||
+                         PyUnicode_Compare(sst, s_closed) == 0)))
+        return kit_scope_goto(S, s_stopped_st);
+    return 0;
+}
+
+static PyObject *
+KitCb_call(PyObject *self_, PyObject *args, PyObject *kwds)
+{
+    KitCb *cb = (KitCb *)self_;
+    SlotKit *k = cb->kc_kit;
+    PyObject *st = NULL;
+    (void)kwds;
+    if (PyTuple_GET_SIZE(args) >= 1)
+        st = PyTuple_GET_ITEM(args, 0);
+    int r = 0;
+    switch (cb->kc_which) {
+    case KCB_BUSY_SMGR:
+        /* record the smgr state as observed through events
+         * (lib/connection-fsm.js:885-890) */
+        if (st != NULL) {
+            Py_INCREF(st);
+            Py_XSETREF(k->k_observed, st);
+        }
+        break;
+    case KCB_BUSY_HDL:
+        if (st == s_released ||
+            (st != NULL && PyUnicode_Compare(st, s_released) == 0))
+            r = kit_busy_on_release(k);
+        else if (st == s_closed ||
+                 (st != NULL && PyUnicode_Compare(st, s_closed) == 0))
+            r = kit_busy_on_close(k);
+        break;
+    case KCB_IDLE_SMGR:
+        r = kit_idle_smgr_changed(k, st);
+        break;
+    case KCB_IDLE_UNWANTED:
+        r = kit_idle_unwanted(k);
+        break;
+    }
+    if (r < 0)
+        return NULL;
+    Py_RETURN_NONE;
+}
+
+static int
+KitCb_traverse(PyObject *self_, visitproc visit, void *arg)
+{
+    Py_VISIT((PyObject *)((KitCb *)self_)->kc_kit);
+    return 0;
+}
+
+static int
+KitCb_clear_(PyObject *self_)
+{
+    Py_CLEAR(((KitCb *)self_)->kc_kit);
+    return 0;
+}
+
+static void
+KitCb_dealloc(PyObject *self_)
+{
+    PyObject_GC_UnTrack(self_);
+    KitCb_clear_(self_);
+    PyObject_GC_Del(self_);
+}
+
+PyTypeObject KitCbType = {
+    PyVarObject_HEAD_INIT(NULL, 0)
+    "cueball_amd._speed._SlotKitCb",
+    sizeof(KitCb),
+};
+
+/* -- the native state entries --------------------------------------- */
+
+static int
+kit_state_busy(SlotKit *k, Scope *scope)
+{
+    FSMOb *slot = k->k_slot;
+    Py_INCREF(t_busy_valid);
+    Py_XSETREF(slot->f_valid, t_busy_valid);
+
+    PyObject *hdl = PyObject_GetAttr((PyObject *)slot, s_csf_handle);
+    if (hdl == NULL)
+        return -1;
+    Py_INCREF(s_connected_st);
+    Py_XSETREF(k->k_observed, s_connected_st);
+    Py_INCREF((PyObject *)scope);
+    Py_XSETREF(k->k_busy_scope, scope);
+
+    if (scope_on_c(scope, k->k_smgr, s_stateChanged,
+                   k->k_cbs[KCB_BUSY_SMGR]) < 0 ||
+        scope_on_c(scope, hdl, s_stateChanged,
+                   k->k_cbs[KCB_BUSY_HDL]) < 0) {
+        Py_DECREF(hdl);
+        return -1;
+    }
+
+    PyObject *sst = ((FSMOb *)k->k_smgr)->f_state;
+    int connected = (sst == s_connected_st ||
+                     (sst != NULL &&
+                      PyUnicode_Compare(sst, s_connected_st) == 0));
+    PyObject *r;
+    if (connected) {
+        PyObject *sock = PyObject_GetAttr(k->k_smgr, s_sm_socket);
+        if (sock == NULL) {
+            Py_DECREF(hdl);
+            return -1;
+        }
+        r = CH_accept(hdl, sock);
+        Py_DECREF(sock);
+        Py_DECREF(hdl);
+        if (r == NULL)
+            return -1;
+        Py_DECREF(r);
+    } else {
+        /* lost the race with the smgr: treat as released
+         * (lib/connection-fsm.js:1129-1196) */
+        r = CH_reject(hdl, NULL);
+        Py_DECREF(hdl);
+        if (r == NULL)
+            return -1;
+        Py_DECREF(r);
+        if (PyObject_SetAttr((PyObject *)slot, s_csf_handle,
+                             Py_None) < 0)
+            return -1;
+        if (kit_busy_on_release(k) < 0)
+            return -1;
+    }
+    return 1;
+}
+
+static int
+kit_state_idle(SlotKit *k, Scope *scope)
+{
+    FSMOb *slot = k->k_slot;
+    /* the monitor-conversion and unwanted cases keep the full Python
+     * entry (cold paths with documented divergences) */
+    int monitor = kit_truthy_attr((PyObject *)slot, s_csf_monitor);
+    if (monitor < 0)
+        return -1;
+    if (monitor)
+        return 0;
+    int wanted = kit_truthy_attr((PyObject *)slot, s_csf_wanted);
+    if (wanted < 0)
+        return -1;
+    if (!wanted)
+        return 0;
+
+    Py_INCREF(t_idle_valid);
+    Py_XSETREF(slot->f_valid, t_idle_valid);
+    Py_INCREF((PyObject *)scope);
+    Py_XSETREF(k->k_idle_scope, scope);
+
+    PyObject *hdl = PyObject_GetAttr((PyObject *)slot, s_csf_handle);
+    if (hdl == NULL)
+        return -1;
+    if (hdl != Py_None) {
+        if (PyObject_SetAttr((PyObject *)slot, s_csf_prev_handle,
+                             hdl) < 0) {
+            Py_DECREF(hdl);
+            return -1;
+        }
+        if (PyObject_SetAttr((PyObject *)slot, s_csf_handle,
+                             Py_None) < 0) {
+            Py_DECREF(hdl);
+            return -1;
+        }
+    }
+    Py_DECREF(hdl);
+
+    if (scope_on_c(scope, k->k_smgr, s_stateChanged,
+                   k->k_cbs[KCB_IDLE_SMGR]) < 0)
+        return -1;
+    if (scope_on_c(scope, (PyObject *)slot, s_unwanted_evt,
+                   k->k_cbs[KCB_IDLE_UNWANTED]) < 0)
+        return -1;
+    return 1;
+}
+
+int
+slotkit_maybe_entry(PyObject *kit_, FSMOb *fsm, PyObject *target,
+                    Scope *scope)
+{
+    SlotKit *k = (SlotKit *)kit_;
+    if (k->k_slot != fsm)
+        return 0;
+    if (target == s_busy_st ||
+        (PyUnicode_Check(target) &&
+         PyUnicode_Compare(target, s_busy_st) == 0))
+        return kit_state_busy(k, scope);
+    if (target == s_idle ||
+        (PyUnicode_Check(target) &&
+         PyUnicode_Compare(target, s_idle) == 0))
+        return kit_state_idle(k, scope);
+    return 0;
+}
+
+/* -- SlotKit type ---------------------------------------------------- */
+
+static int
+SlotKit_traverse(PyObject *self_, visitproc visit, void *arg)
+{
+    SlotKit *k = (SlotKit *)self_;
+    Py_VISIT((PyObject *)k->k_slot);
+    Py_VISIT(k->k_smgr);
+    Py_VISIT(k->k_observed);
+    Py_VISIT((PyObject *)k->k_busy_scope);
+    Py_VISIT((PyObject *)k->k_idle_scope);
+    for (int i = 0; i < 4; i++)
+        Py_VISIT(k->k_cbs[i]);
+    return 0;
+}
+
+static int
+SlotKit_clear_(PyObject *self_)
+{
+    SlotKit *k = (SlotKit *)self_;
+    Py_CLEAR(k->k_slot);
+    Py_CLEAR(k->k_smgr);
+    Py_CLEAR(k->k_observed);
+    Py_CLEAR(k->k_busy_scope);
+    Py_CLEAR(k->k_idle_scope);
+    for (int i = 0; i < 4; i++)
+        Py_CLEAR(k->k_cbs[i]);
+    return 0;
+}
+
+static void
+SlotKit_dealloc(PyObject *self_)
+{
+    PyObject_GC_UnTrack(self_);
+    SlotKit_clear_(self_);
+    PyObject_GC_Del(self_);
+}
+
+static PyObject *
+SlotKit_new(PyTypeObject *type, PyObject *args, PyObject *kwds)
+{
+    PyObject *slot, *smgr;
+    (void)kwds;
+    if (!PyArg_ParseTuple(args, "OO", &slot, &smgr))
+        return NULL;
+    if (!PyObject_TypeCheck(slot, &FSMType) ||
+        !PyObject_TypeCheck(smgr, &FSMType)) {
+        PyErr_SetString(PyExc_TypeError,
+                        "SlotKit(slot, smgr): native FSMs required");
+        return NULL;
+    }
+    SlotKit *k = PyObject_GC_New(SlotKit, type);
+    if (k == NULL)
+        return NULL;
+    Py_INCREF(slot);
+    k->k_slot = (FSMOb *)slot;
+    Py_INCREF(smgr);
+    k->k_smgr = smgr;
+    k->k_observed = NULL;
+    k->k_busy_scope = NULL;
+    k->k_idle_scope = NULL;
+    for (int i = 0; i < 4; i++)
+        k->k_cbs[i] = NULL;
+    PyObject_GC_Track((PyObject *)k);
+    for (int i = 0; i < 4; i++) {
+        KitCb *cb = PyObject_GC_New(KitCb, &KitCbType);
+        if (cb == NULL) {
+            Py_DECREF((PyObject *)k);
+            return NULL;
+        }
+        Py_INCREF((PyObject *)k);
+        cb->kc_kit = k;
+        cb->kc_which = i;
+        PyObject_GC_Track((PyObject *)cb);
+        k->k_cbs[i] = (PyObject *)cb;
+    }
+    return (PyObject *)k;
+}
+
+PyTypeObject SlotKitType = {
+    PyVarObject_HEAD_INIT(NULL, 0)
+    "cueball_amd._speed.SlotKit",
+    sizeof(SlotKit),
+};
+
+static void
+slotkit_types_init(void)
+{
+    KitCbType.tp_dealloc = KitCb_dealloc;
+    KitCbType.tp_call = KitCb_call;
+    KitCbType.tp_flags = Py_TPFLAGS_DEFAULT | Py_TPFLAGS_HAVE_GC;
+    KitCbType.tp_traverse = KitCb_traverse;
+    KitCbType.tp_clear = KitCb_clear_;
+
+    SlotKitType.tp_dealloc = SlotKit_dealloc;
+    SlotKitType.tp_flags = Py_TPFLAGS_DEFAULT | Py_TPFLAGS_HAVE_GC;
+    SlotKitType.tp_traverse = SlotKit_traverse;
+    SlotKitType.tp_clear = SlotKit_clear_;
+    SlotKitType.tp_new = SlotKit_new;
+}
+
 PyObject *
 speed_set_tracer(PyObject *mod, PyObject *fn)
 {
@@ -3283,8 +3776,27 @@ PyInit__speed(void)
     s_p_busy_hwm = PyUnicode_InternFromString("p_busy_hwm");
     s_p_demand_hwm = PyUnicode_InternFromString("p_demand_hwm");
     s_ticket_slow = PyUnicode_InternFromString("_ticket_slow");
+    s_connected_st = PyUnicode_InternFromString("connected");
+    s_busy_st = PyUnicode_InternFromString("busy");
+    s_stopping_st = PyUnicode_InternFromString("stopping");
+    s_stopped_st = PyUnicode_InternFromString("stopped");
+    s_retrying_st = PyUnicode_InternFromString("retrying");
+    s_connecting_st = PyUnicode_InternFromString("connecting");
+    s_killing_st = PyUnicode_InternFromString("killing");
+    s_unwanted_evt = PyUnicode_InternFromString("unwanted");
+    s_csf_handle = PyUnicode_InternFromString("csf_handle");
+    s_csf_prev_handle = PyUnicode_InternFromString("csf_prev_handle");
+    s_csf_wanted = PyUnicode_InternFromString("csf_wanted");
+    s_csf_monitor = PyUnicode_InternFromString("csf_monitor");
+    s_sm_socket = PyUnicode_InternFromString("sm_socket");
+    t_busy_valid = PyTuple_Pack(6, s_idle, s_stopping_st, s_stopped_st,
+                                s_retrying_st, s_killing_st,
+                                s_connecting_st);
+    t_idle_valid = PyTuple_Pack(5, s_retrying_st, s_connecting_st,
+                                s_stopping_st, s_stopped_st, s_busy_st);
     queue_types_init();
     ct_type_init();
+    slotkit_types_init();
     if (PyType_Ready(&EmitterType) < 0 ||
         PyType_Ready(&OnceWrapperType) < 0 ||
         PyType_Ready(&GuardedCbType) < 0 ||
@@ -3297,7 +3809,9 @@ PyInit__speed(void)
         PyType_Ready(&FSMType) < 0 ||
         PyType_Ready(&QNodeType) < 0 ||
         PyType_Ready(&NQueueType) < 0 ||
-        PyType_Ready(&CTType) < 0)
+        PyType_Ready(&CTType) < 0 ||
+        PyType_Ready(&KitCbType) < 0 ||
+        PyType_Ready(&SlotKitType) < 0)
         return NULL;
 
     g_remove_desc = PyDict_GetItemString(EmitterType.tp_dict,
@@ -3323,5 +3837,7 @@ PyInit__speed(void)
     PyModule_AddObject(m, "QueueNode", (PyObject *)&QNodeType);
     Py_INCREF(&CTType);
     PyModule_AddObject(m, "ClaimTicket", (PyObject *)&CTType);
+    Py_INCREF(&SlotKitType);
+    PyModule_AddObject(m, "SlotKit", (PyObject *)&SlotKitType);
     return m;
 }

@@ -48,10 +48,13 @@ try:
     import os as _os
     if _os.environ.get("CUEBALL_PURE"):
         _native_count = None
+        _SlotKit = None
     else:
         from ._speed import count_listeners as _native_count
+        from ._speed import SlotKit as _SlotKit
 except ImportError:
     _native_count = None
+    _SlotKit = None
 
 
 def _call_optional(obj: Any, *names: str) -> None:
@@ -649,6 +652,14 @@ class ConnectionSlotFSM(FSM):
             "loop": options.get("loop"),
         })
         super().__init__("init", loop=options.get("loop"))
+        # Native busy/idle hot cycle: with the C core loaded and no
+        # health checker configured, state_busy/state_idle and their
+        # event callbacks run natively (zero Python allocation per
+        # claim cycle); monitor/unwanted idle entries and every other
+        # state still use the Python entries above.
+        if _SlotKit is not None and self.csf_checker is None:
+            self._csf_kit = _SlotKit(self, self.csf_smgr)
+            self._set_fast_kit(self._csf_kit)
 
     # -- signal functions ----------------------------------------------
     def set_unwanted(self) -> None:
