@@ -722,39 +722,42 @@ class ConnectionPool(FSM):
                 del self.p_dead[key]
                 self.rebalance()
 
-        if new_state == "idle" and fsm.is_in_state("idle"):
-            # Just became available: either released by its user or done
-            # connecting.
-            if key not in self.p_backends:
-                fsm.set_unwanted()
+            if fsm._fsm_state == "idle":
+                # Just became available: either released by its user
+                # or done connecting.
+                if key not in self.p_backends:
+                    fsm.set_unwanted()
+                    return
+
+                # Feed waiters, with the CoDel drop check on each.
+                while self.p_waiters._len > 0:
+                    hdl = self.p_waiters.shift()
+                    drop = self.p_codel is not None and \
+                        self.p_codel.overloaded(hdl.ch_started)
+                    if not hdl.is_in_state("waiting"):
+                        continue
+                    if drop:
+                        hdl.timeout()
+                        continue
+                    hdl.try_(fsm)
+                    return
+
+                if self.p_codel is not None:
+                    self.p_codel.empty()
+
+                fsm.p_idleq_node = self.p_idleq.push(fsm)
                 return
 
-            # Feed waiters, with the CoDel drop check on each.
-            while self.p_waiters._len > 0:
-                hdl = self.p_waiters.shift()
-                drop = self.p_codel is not None and \
-                    self.p_codel.overloaded(hdl.ch_started)
-                if not hdl.is_in_state("waiting"):
-                    continue
-                if drop:
-                    hdl.timeout()
-                    continue
-                hdl.try_(fsm)
-                return
+        elif new_state == "busy":
+            # Health-checking connections ride the initq so they don't
+            # count as busy (lib/pool.js:762-769); inline of
+            # fsm.is_running_ping() with the cheap checks first
+            hdl = fsm.csf_handle
+            if hdl is not None and fsm.p_initq_node is None and \
+                    hdl.ch_pinger and fsm._fsm_state == "busy":
+                fsm.p_initq_node = self.p_initq.push(fsm)
 
-            if self.p_codel is not None:
-                self.p_codel.empty()
-
-            fsm.p_idleq_node = self.p_idleq.push(fsm)
-            return
-
-        # Health-checking connections ride the initq so they don't count
-        # as busy (lib/pool.js:762-769).
-        if new_state == "busy" and fsm.is_running_ping() and \
-                fsm.p_initq_node is None:
-            fsm.p_initq_node = self.p_initq.push(fsm)
-
-        if new_state == "failed":
+        elif new_state == "failed":
             if key in self.p_backends:
                 self.p_dead[key] = True
             err = fsm.get_socket_mgr().get_last_error()
@@ -854,30 +857,39 @@ class ConnectionPool(FSM):
         Returns the ClaimHandle (or a cancel-only stub when the pool is
         stopping/failed, lib/pool.js:889-910).
         """
-        if callable(options) and cb is None:
-            cb = options
-            options = {}
-        options = options or {}
-        if cb is None:
-            raise TypeError("claim() requires a callback")
-        err_on_empty = bool(options.get("errorOnEmpty", False))
-
-        if self.p_codel is not None:
-            if options.get("timeout") is not None:
-                raise ValueError("options.timeout not allowed when "
-                                 "targetClaimDelay has been set")
-            timeout = self.p_codel.get_max_idle()
-        elif options.get("timeout") is not None:
-            timeout = options["timeout"]
-        else:
+        # fast path: claim({}, cb) / claim(cb) with no CoDel — the
+        # overwhelmingly common call shape on the hot path
+        if not options and cb is not None and self.p_codel is None:
+            err_on_empty = False
             timeout = math.inf
+        else:
+            if callable(options) and cb is None:
+                cb = options
+                options = {}
+            options = options or {}
+            if cb is None:
+                raise TypeError("claim() requires a callback")
+            err_on_empty = bool(options.get("errorOnEmpty", False))
 
-        self._incr_counter("claim")
+            if self.p_codel is not None:
+                if options.get("timeout") is not None:
+                    raise ValueError("options.timeout not allowed when "
+                                     "targetClaimDelay has been set")
+                timeout = self.p_codel.get_max_idle()
+            elif options.get("timeout") is not None:
+                timeout = options["timeout"]
+            else:
+                timeout = math.inf
 
-        if self.is_in_state("stopping") or self.is_in_state("stopped"):
+        counters = self.p_counters
+        counters["claim"] = counters.get("claim", 0) + 1
+
+        st = self._fsm_state
+        if st == "stopping" or st == "stopped" or \
+                st == "stopping.backends":
             return self._claim_shortcircuit(
                 cb, mod_errors.PoolStoppingError(self))
-        if self.is_in_state("failed"):
+        if st == "failed":
             return self._claim_shortcircuit(
                 cb, mod_errors.PoolFailedError(self, self.p_last_error))
 
