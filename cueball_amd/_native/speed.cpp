@@ -3252,6 +3252,94 @@ PyTypeObject CTType = {
     sizeof(CTOb),
 };
 
+/* claim_fast(handle_cls, pool, stack, cb, log, timeout, loop,
+ *            err_on_empty) -> handle
+ * One C call for pool.claim's hot tail: allocate the ClaimHandle
+ * subclass, run the _setup core (throw_error=True), build the native
+ * ticket and register it as the stateChanged listener. */
+PyObject *CH__setup(PyObject *self_, PyObject *const *args,
+                    Py_ssize_t nargs);
+
+static PyObject *
+speed_claim_fast(PyObject *mod, PyObject *const *args, Py_ssize_t nargs)
+{
+    (void)mod;
+    if (nargs != 8) {
+        PyErr_SetString(PyExc_TypeError,
+            "claim_fast(cls, pool, stack, cb, log, timeout, loop, "
+            "err_on_empty)");
+        return NULL;
+    }
+    PyTypeObject *cls = (PyTypeObject *)args[0];
+    if (!PyType_Check(args[0]) ||
+        !PyType_IsSubtype(cls, &CHType)) {
+        PyErr_SetString(PyExc_TypeError,
+                        "cls must be a ClaimHandleBase subclass");
+        return NULL;
+    }
+    PyObject *pool = args[1];
+    if (!PyObject_TypeCheck(pool, &FSMType)) {
+        PyErr_SetString(PyExc_TypeError, "pool must be a native FSM");
+        return NULL;
+    }
+    int err_on_empty = PyObject_IsTrue(args[7]);
+    if (err_on_empty < 0)
+        return NULL;
+
+    PyObject *handle = cls->tp_alloc(cls, 0);
+    if (handle == NULL)
+        return NULL;
+    PyObject *setup_args[7] = {pool, args[2], args[3], args[4],
+                               args[5], Py_True, args[6]};
+    PyObject *r = CH__setup(handle, setup_args, 7);
+    if (r == NULL) {
+        Py_DECREF(handle);
+        return NULL;
+    }
+    Py_DECREF(r);
+
+    CTOb *t = PyObject_GC_New(CTOb, &CTType);
+    if (t == NULL) {
+        Py_DECREF(handle);
+        return NULL;
+    }
+    Py_INCREF(pool);
+    t->ct_pool = pool;
+    Py_INCREF(handle);
+    t->ct_handle = handle;
+    t->ct_idleq = NULL;
+    t->ct_waiters = NULL;
+    t->ct_initq = NULL;
+    t->ct_err_on_empty = err_on_empty;
+    PyObject_GC_Track((PyObject *)t);
+    PyObject *iq = PyObject_GetAttrString(pool, "p_idleq");
+    PyObject *wq = iq ? PyObject_GetAttrString(pool, "p_waiters") : NULL;
+    PyObject *nq = wq ? PyObject_GetAttrString(pool, "p_initq") : NULL;
+    if (nq == NULL) {
+        PyErr_Clear();
+        Py_XDECREF(iq);
+        Py_XDECREF(wq);
+    } else if (PyObject_TypeCheck(iq, &NQueueType) &&
+               PyObject_TypeCheck(wq, &NQueueType) &&
+               PyObject_TypeCheck(nq, &NQueueType)) {
+        t->ct_idleq = (NQueue *)iq;
+        t->ct_waiters = (NQueue *)wq;
+        t->ct_initq = (NQueue *)nq;
+    } else {
+        Py_DECREF(iq);
+        Py_DECREF(wq);
+        Py_DECREF(nq);
+    }
+    r = emitter_add((Emitter *)handle, s_stateChanged, (PyObject *)t);
+    Py_DECREF((PyObject *)t);
+    if (r == NULL) {
+        Py_DECREF(handle);
+        return NULL;
+    }
+    Py_DECREF(r);
+    return handle;
+}
+
 static void
 ct_type_init(void)
 {
@@ -3710,6 +3798,8 @@ PyMethodDef speed_methods[] = {
      (PyCFunction)(void (*)(void))speed_count_listeners, METH_FASTCALL,
      NULL},
     {"_set_helpers", (PyCFunction)(void (*)(void))speed_set_helpers,
+     METH_FASTCALL, NULL},
+    {"claim_fast", (PyCFunction)(void (*)(void))speed_claim_fast,
      METH_FASTCALL, NULL},
     {NULL, NULL, 0, NULL},
 };

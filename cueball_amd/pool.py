@@ -105,9 +105,11 @@ class _ClaimTicket:
 
 
 _NativeClaimTicket = None
+_native_claim_fast = None
 if not _os.environ.get("CUEBALL_PURE"):
     try:
         from ._speed import ClaimTicket as _NativeClaimTicket  # noqa: F811
+        from ._speed import claim_fast as _native_claim_fast
     except ImportError:
         pass
 
@@ -894,17 +896,17 @@ class ConnectionPool(FSM):
                 cb, mod_errors.PoolFailedError(self, self.p_last_error))
 
         stack = mod_utils.maybe_capture_stack_trace()
-        handle = ClaimHandle.fast(self, stack, cb, self.p_claim_log,
-                                  timeout, self._loop)
-        # The handle's construction already queued its async
+        # The handle's construction already queues its async
         # stateChanged('waiting'); the ticket receives it on the next
         # loop turn and runs the first try_next then — claim() never
         # fires the callback synchronously (lib/pool.js:922-968).
-        if _NativeClaimTicket is not None:
-            ticket = _NativeClaimTicket(self, handle, err_on_empty)
-        else:
-            ticket = _ClaimTicket(self, handle, err_on_empty)
-        handle.on("stateChanged", ticket)
+        if _native_claim_fast is not None:
+            return _native_claim_fast(ClaimHandle, self, stack, cb,
+                                      self.p_claim_log, timeout,
+                                      self._loop, err_on_empty)
+        handle = ClaimHandle.fast(self, stack, cb, self.p_claim_log,
+                                  timeout, self._loop)
+        handle.on("stateChanged", _ClaimTicket(self, handle, err_on_empty))
         return handle
 
     def _claim_shortcircuit(self, cb: Callable, err: BaseException):
