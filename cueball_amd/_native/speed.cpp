@@ -1697,6 +1697,10 @@ PyObject *t_claimed_valid;       /* ("released","closed") */
 PyObject *t_empty;               /* () */
 PyObject *s_leak_events[4];      /* close, error, readable, data */
 
+/* defined in the SlotKit section below */
+extern PyObject *s_busy_st;
+extern PyObject *s_csf_handle;
+
 typedef struct {
     FSMOb base;
     PyObject *chb_pool;
@@ -2173,7 +2177,34 @@ CH_state_claiming(PyObject *self_, PyObject *scope)
     (void)scope;
     Py_INCREF(t_claiming_valid);
     Py_XSETREF(self->base.f_valid, t_claiming_valid);
-    PyObject *r = PyObject_CallMethodObjArgs(self->chb_slot, s_claim,
+    PyObject *slot = self->chb_slot;
+    /* inline of ConnectionSlotFSM.claim() when the slot runs the
+     * native busy/idle cycle (same checks, same transition) */
+    if (slot != NULL && PyObject_TypeCheck(slot, &FSMType) &&
+        ((FSMOb *)slot)->f_fastkit != NULL) {
+        FSMOb *sf = (FSMOb *)slot;
+        PyObject *fs = sf->f_state;
+        if (!(fs == s_idle ||
+              (fs != NULL && PyUnicode_Compare(fs, s_idle) == 0))) {
+            PyErr_SetString(g_fsm_error, "claim only in idle");
+            return NULL;
+        }
+        PyObject *cur = PyObject_GetAttr(slot, s_csf_handle);
+        if (cur == NULL)
+            return NULL;
+        int has = (cur != Py_None);
+        Py_DECREF(cur);
+        if (has) {
+            PyErr_SetString(g_fsm_error, "slot already has a handle");
+            return NULL;
+        }
+        if (PyObject_SetAttr(slot, s_csf_handle, self_) < 0)
+            return NULL;
+        if (fsm_goto_state(sf, s_busy_st) < 0)
+            return NULL;
+        Py_RETURN_NONE;
+    }
+    PyObject *r = PyObject_CallMethodObjArgs(slot, s_claim,
                                              self_, NULL);
     if (r == NULL)
         return NULL;
@@ -3675,6 +3706,285 @@ slotkit_maybe_entry(PyObject *kit_, FSMOb *fsm, PyObject *target,
     return 0;
 }
 
+/* ------------------------------------------------------------------ */
+/* SlotDispatch: the pool's per-slot stateChanged listener             */
+/* (pool._slot_state_changed) with the idle-feed and busy no-op fast   */
+/* paths in C; every other state (and every unusual condition)         */
+/* delegates to the Python method, which remains the full logic.       */
+
+PyObject *s_p_backends;          /* "p_backends" */
+PyObject *s_p_dead;              /* "p_dead" */
+PyObject *s_p_initq_node;        /* "p_initq_node" */
+PyObject *s_connected_to_backend;/* "connectedToBackend" */
+PyObject *s_slot_state_changed;  /* "_slot_state_changed" */
+PyObject *s_remove_node;         /* "remove" */
+PyObject *s_rebalance;           /* "rebalance" */
+PyObject *s_set_unwanted;        /* "set_unwanted" */
+
+typedef struct {
+    PyObject_HEAD
+    PyObject *sd_pool;      /* owned */
+    PyObject *sd_fsm;       /* the slot (FSMOb), owned */
+    PyObject *sd_key;       /* owned */
+    NQueue *sd_idleq;       /* owned (never replaced by the pool) */
+    NQueue *sd_waiters;     /* owned */
+    int sd_has_codel;       /* CoDel pools use the python path */
+} SlotDispatch;
+
+extern PyTypeObject SlotDispatchType;
+
+static int
+sd_fallback(SlotDispatch *d, PyObject *st)
+{
+    PyObject *r = PyObject_CallMethodObjArgs(
+        d->sd_pool, s_slot_state_changed, d->sd_fsm, d->sd_key, st,
+        NULL);
+    if (r == NULL)
+        return -1;
+    Py_DECREF(r);
+    return 0;
+}
+
+static PyObject *
+SlotDispatch_call(PyObject *self_, PyObject *args, PyObject *kwds)
+{
+    SlotDispatch *d = (SlotDispatch *)self_;
+    (void)kwds;
+    if (PyTuple_GET_SIZE(args) != 1) {
+        PyErr_SetString(PyExc_TypeError, "SlotDispatch(state)");
+        return NULL;
+    }
+    PyObject *st = PyTuple_GET_ITEM(args, 0);
+    FSMOb *fsm = (FSMOb *)d->sd_fsm;
+
+    /* common preconditions for both fast paths */
+    PyObject *initq_node = PyObject_GetAttr(d->sd_fsm, s_p_initq_node);
+    if (initq_node == NULL)
+        return NULL;
+    int in_initq = (initq_node != Py_None);
+    Py_DECREF(initq_node);
+
+    if (!in_initq &&
+        (st == s_busy_st ||
+         (PyUnicode_Check(st) &&
+          PyUnicode_Compare(st, s_busy_st) == 0))) {
+        /* busy: with no pinger handle and no stale idleq node this
+         * dispatch is a no-op */
+        PyObject *idleq_node = PyObject_GetAttr(d->sd_fsm,
+                                                s_p_idleq_node);
+        if (idleq_node == NULL)
+            return NULL;
+        int linked = (idleq_node != Py_None);
+        Py_DECREF(idleq_node);
+        if (!linked) {
+            PyObject *hdl = PyObject_GetAttr(d->sd_fsm, s_csf_handle);
+            if (hdl == NULL)
+                return NULL;
+            int pinger = 0;
+            if (hdl != Py_None && PyObject_TypeCheck(hdl, &CHType))
+                pinger = ((CHOb *)hdl)->chb_pinger;
+            Py_DECREF(hdl);
+            if (!pinger)
+                Py_RETURN_NONE;
+        }
+        if (sd_fallback(d, st) < 0)
+            return NULL;
+        Py_RETURN_NONE;
+    }
+
+    if (!in_initq && !d->sd_has_codel &&
+        (st == s_idle ||
+         (PyUnicode_Check(st) &&
+          PyUnicode_Compare(st, s_idle) == 0))) {
+        PyObject *dead = PyObject_GetAttr(d->sd_pool, s_p_dead);
+        if (dead == NULL)
+            return NULL;
+        int isdead = PyDict_Check(dead) ?
+            PyDict_Contains(dead, d->sd_key) : 1;
+        Py_DECREF(dead);
+        if (isdead != 0) {
+            /* dead-backend recovery (or error): python path */
+            if (isdead < 0)
+                return NULL;
+            if (sd_fallback(d, st) < 0)
+                return NULL;
+            Py_RETURN_NONE;
+        }
+        /* emit connectedToBackend (cheap when nobody listens) */
+        PyObject *eargs[3] = {s_connected_to_backend, d->sd_key,
+                              d->sd_fsm};
+        if (emitter_emit_core((Emitter *)d->sd_pool, eargs[0],
+                              eargs + 1, 2) < 0)
+            return NULL;
+        /* still idle? (stale events tolerated) */
+        PyObject *fs = fsm->f_state;
+        if (!(fs == s_idle ||
+              (fs != NULL && PyUnicode_Compare(fs, s_idle) == 0))) {
+            /* the event already emitted above; only the trailing
+             * was-idle-now-isn't unlink branch can still apply */
+            PyObject *idn = PyObject_GetAttr(d->sd_fsm,
+                                             s_p_idleq_node);
+            if (idn == NULL)
+                return NULL;
+            if (idn != Py_None) {
+                PyObject *rr = PyObject_CallMethodObjArgs(
+                    idn, s_remove_node, NULL);
+                if (rr == NULL) {
+                    Py_DECREF(idn);
+                    return NULL;
+                }
+                Py_DECREF(rr);
+                Py_DECREF(idn);
+                if (PyObject_SetAttr(d->sd_fsm, s_p_idleq_node,
+                                     Py_None) < 0)
+                    return NULL;
+                PyObject *rb = PyObject_CallMethodObjArgs(
+                    d->sd_pool, s_rebalance, NULL);
+                if (rb == NULL)
+                    return NULL;
+                Py_DECREF(rb);
+            } else {
+                Py_DECREF(idn);
+            }
+            Py_RETURN_NONE;
+        }
+        {
+            PyObject *backends = PyObject_GetAttr(d->sd_pool,
+                                                  s_p_backends);
+            if (backends == NULL)
+                return NULL;
+            int wanted = PyDict_Check(backends) ?
+                PyDict_Contains(backends, d->sd_key) : 0;
+            Py_DECREF(backends);
+            if (wanted < 0)
+                return NULL;
+            if (!wanted) {
+                /* no longer a backend: set_unwanted, like the python
+                 * dispatcher (it has not emitted again: call the slot
+                 * method directly) */
+                PyObject *su = PyObject_CallMethodObjArgs(
+                    d->sd_fsm, s_set_unwanted, NULL);
+                if (su == NULL)
+                    return NULL;
+                Py_DECREF(su);
+                Py_RETURN_NONE;
+            }
+        }
+        /* feed waiters */
+        while (d->sd_waiters->q_len > 0) {
+            PyObject *hdl = nqueue_shift_value(d->sd_waiters);
+            if (hdl == NULL)
+                return NULL;
+            if (!PyObject_TypeCheck(hdl, &CHType) ||
+                !ch_state_is((CHOb *)hdl, s_waiting)) {
+                Py_DECREF(hdl);
+                continue;
+            }
+            PyObject *r = CH_try_(hdl, d->sd_fsm);
+            Py_DECREF(hdl);
+            if (r == NULL)
+                return NULL;
+            Py_DECREF(r);
+            Py_RETURN_NONE;
+        }
+        /* no waiter: park on the idle queue */
+        QNode *n = nqueue_push(d->sd_idleq, d->sd_fsm);
+        if (n == NULL)
+            return NULL;
+        int rc = PyObject_SetAttr(d->sd_fsm, s_p_idleq_node,
+                                  (PyObject *)n);
+        Py_DECREF((PyObject *)n);
+        if (rc < 0)
+            return NULL;
+        Py_RETURN_NONE;
+    }
+
+    if (sd_fallback(d, st) < 0)
+        return NULL;
+    Py_RETURN_NONE;
+}
+
+static int
+SlotDispatch_traverse(PyObject *self_, visitproc visit, void *arg)
+{
+    SlotDispatch *d = (SlotDispatch *)self_;
+    Py_VISIT(d->sd_pool);
+    Py_VISIT(d->sd_fsm);
+    Py_VISIT(d->sd_key);
+    Py_VISIT((PyObject *)d->sd_idleq);
+    Py_VISIT((PyObject *)d->sd_waiters);
+    return 0;
+}
+
+static int
+SlotDispatch_clear_(PyObject *self_)
+{
+    SlotDispatch *d = (SlotDispatch *)self_;
+    Py_CLEAR(d->sd_pool);
+    Py_CLEAR(d->sd_fsm);
+    Py_CLEAR(d->sd_key);
+    Py_CLEAR(d->sd_idleq);
+    Py_CLEAR(d->sd_waiters);
+    return 0;
+}
+
+static void
+SlotDispatch_dealloc(PyObject *self_)
+{
+    PyObject_GC_UnTrack(self_);
+    SlotDispatch_clear_(self_);
+    PyObject_GC_Del(self_);
+}
+
+static PyObject *
+SlotDispatch_new(PyTypeObject *type, PyObject *args, PyObject *kwds)
+{
+    PyObject *pool, *fsm, *key, *has_codel;
+    (void)kwds;
+    if (!PyArg_ParseTuple(args, "OOOO", &pool, &fsm, &key, &has_codel))
+        return NULL;
+    if (!PyObject_TypeCheck(pool, &EmitterType) ||
+        !PyObject_TypeCheck(fsm, &FSMType)) {
+        PyErr_SetString(PyExc_TypeError,
+                        "SlotDispatch(pool, slot, key, has_codel)");
+        return NULL;
+    }
+    PyObject *iq = PyObject_GetAttrString(pool, "p_idleq");
+    PyObject *wq = iq ? PyObject_GetAttrString(pool, "p_waiters") : NULL;
+    if (wq == NULL || !PyObject_TypeCheck(iq, &NQueueType) ||
+        !PyObject_TypeCheck(wq, &NQueueType)) {
+        Py_XDECREF(iq);
+        Py_XDECREF(wq);
+        if (!PyErr_Occurred())
+            PyErr_SetString(PyExc_TypeError,
+                            "pool queues must be native");
+        return NULL;
+    }
+    SlotDispatch *d = PyObject_GC_New(SlotDispatch, type);
+    if (d == NULL) {
+        Py_DECREF(iq);
+        Py_DECREF(wq);
+        return NULL;
+    }
+    Py_INCREF(pool);
+    d->sd_pool = pool;
+    Py_INCREF(fsm);
+    d->sd_fsm = fsm;
+    Py_INCREF(key);
+    d->sd_key = key;
+    d->sd_idleq = (NQueue *)iq;
+    d->sd_waiters = (NQueue *)wq;
+    d->sd_has_codel = PyObject_IsTrue(has_codel);
+    PyObject_GC_Track((PyObject *)d);
+    return (PyObject *)d;
+}
+
+PyTypeObject SlotDispatchType = {
+    PyVarObject_HEAD_INIT(NULL, 0)
+    "cueball_amd._speed.SlotDispatch",
+    sizeof(SlotDispatch),
+};
+
 /* -- SlotKit type ---------------------------------------------------- */
 
 static int
@@ -3774,6 +4084,13 @@ slotkit_types_init(void)
     SlotKitType.tp_traverse = SlotKit_traverse;
     SlotKitType.tp_clear = SlotKit_clear_;
     SlotKitType.tp_new = SlotKit_new;
+
+    SlotDispatchType.tp_dealloc = SlotDispatch_dealloc;
+    SlotDispatchType.tp_call = SlotDispatch_call;
+    SlotDispatchType.tp_flags = Py_TPFLAGS_DEFAULT | Py_TPFLAGS_HAVE_GC;
+    SlotDispatchType.tp_traverse = SlotDispatch_traverse;
+    SlotDispatchType.tp_clear = SlotDispatch_clear_;
+    SlotDispatchType.tp_new = SlotDispatch_new;
 }
 
 PyObject *
@@ -3884,6 +4201,16 @@ PyInit__speed(void)
                                 s_connecting_st);
     t_idle_valid = PyTuple_Pack(5, s_retrying_st, s_connecting_st,
                                 s_stopping_st, s_stopped_st, s_busy_st);
+    s_p_backends = PyUnicode_InternFromString("p_backends");
+    s_p_dead = PyUnicode_InternFromString("p_dead");
+    s_p_initq_node = PyUnicode_InternFromString("p_initq_node");
+    s_connected_to_backend =
+        PyUnicode_InternFromString("connectedToBackend");
+    s_slot_state_changed =
+        PyUnicode_InternFromString("_slot_state_changed");
+    s_remove_node = PyUnicode_InternFromString("remove");
+    s_rebalance = PyUnicode_InternFromString("rebalance");
+    s_set_unwanted = PyUnicode_InternFromString("set_unwanted");
     queue_types_init();
     ct_type_init();
     slotkit_types_init();
@@ -3901,7 +4228,8 @@ PyInit__speed(void)
         PyType_Ready(&NQueueType) < 0 ||
         PyType_Ready(&CTType) < 0 ||
         PyType_Ready(&KitCbType) < 0 ||
-        PyType_Ready(&SlotKitType) < 0)
+        PyType_Ready(&SlotKitType) < 0 ||
+        PyType_Ready(&SlotDispatchType) < 0)
         return NULL;
 
     g_remove_desc = PyDict_GetItemString(EmitterType.tp_dict,
@@ -3929,5 +4257,7 @@ PyInit__speed(void)
     PyModule_AddObject(m, "ClaimTicket", (PyObject *)&CTType);
     Py_INCREF(&SlotKitType);
     PyModule_AddObject(m, "SlotKit", (PyObject *)&SlotKitType);
+    Py_INCREF(&SlotDispatchType);
+    PyModule_AddObject(m, "SlotDispatch", (PyObject *)&SlotDispatchType);
     return m;
 }

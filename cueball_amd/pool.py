@@ -106,10 +106,12 @@ class _ClaimTicket:
 
 _NativeClaimTicket = None
 _native_claim_fast = None
+_NativeSlotDispatch = None
 if not _os.environ.get("CUEBALL_PURE"):
     try:
         from ._speed import ClaimTicket as _NativeClaimTicket  # noqa: F811
         from ._speed import claim_fast as _native_claim_fast
+        from ._speed import SlotDispatch as _NativeSlotDispatch
     except ImportError:
         pass
 
@@ -703,10 +705,15 @@ class ConnectionPool(FSM):
         fsm.p_initq_node = self.p_initq.push(fsm)
         fsm.p_idleq_node = None
 
-        # functools.partial dispatches at C level (the per-event lambda
-        # wrapper showed up in the claim-path profile)
-        fsm.on("stateChanged",
-               functools.partial(self._slot_state_changed, fsm, key))
+        # Native dispatcher when available (idle-feed/busy fast paths
+        # in C, everything else falls back to _slot_state_changed);
+        # else functools.partial, which dispatches at C level.
+        if _NativeSlotDispatch is not None:
+            fsm.on("stateChanged", _NativeSlotDispatch(
+                self, fsm, key, self.p_codel is not None))
+        else:
+            fsm.on("stateChanged",
+                   functools.partial(self._slot_state_changed, fsm, key))
         fsm.start()
 
     def _slot_state_changed(self, fsm: ConnectionSlotFSM, key: str,
