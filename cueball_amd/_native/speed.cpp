@@ -3103,6 +3103,100 @@ ct_note_demand(CTOb *t)
     return 0;
 }
 
+PyObject *s_p_counters;          /* "p_counters" */
+PyObject *s_p_rebal_scheduled;   /* "p_rebal_scheduled" */
+PyObject *s_max_claim_queue;     /* "max-claim-queue" */
+PyObject *s_queued_claim;        /* "queued-claim" */
+extern PyObject *s_rebalance;    /* "rebalance" (SlotDispatch section) */
+
+/* the waiter-enqueue branch of pool._ticket_slow in C: push the
+ * handle, fold the demand sample, bump the HWM/queued counters, and
+ * schedule a rebalance unless one is already pending */
+static int
+ct_enqueue_waiter(CTOb *t)
+{
+    QNode *n = nqueue_push(t->ct_waiters, t->ct_handle);
+    if (n == NULL)
+        return -1;
+    Py_DECREF((PyObject *)n);
+    if (ct_note_demand(t) < 0)
+        return -1;
+
+    PyObject *counters = PyObject_GetAttr(t->ct_pool, s_p_counters);
+    if (counters == NULL)
+        return -1;
+    if (!PyDict_Check(counters)) {
+        Py_DECREF(counters);
+        PyErr_SetString(PyExc_TypeError, "p_counters must be a dict");
+        return -1;
+    }
+    /* max-claim-queue HWM */
+    PyObject *cur = PyDict_GetItemWithError(counters, s_max_claim_queue);
+    if (cur == NULL && PyErr_Occurred()) {
+        Py_DECREF(counters);
+        return -1;
+    }
+    long curv = -1;
+    if (cur != NULL) {
+        curv = PyLong_AsLong(cur);
+        if (curv == -1 && PyErr_Occurred()) {
+            Py_DECREF(counters);
+            return -1;
+        }
+    }
+    if ((long)t->ct_waiters->q_len > curv) {
+        PyObject *nv = PyLong_FromSsize_t(t->ct_waiters->q_len);
+        if (nv == NULL || PyDict_SetItem(counters, s_max_claim_queue,
+                                         nv) < 0) {
+            Py_XDECREF(nv);
+            Py_DECREF(counters);
+            return -1;
+        }
+        Py_DECREF(nv);
+    }
+    /* queued-claim += 1 (not a tracked metric event) */
+    cur = PyDict_GetItemWithError(counters, s_queued_claim);
+    if (cur == NULL && PyErr_Occurred()) {
+        Py_DECREF(counters);
+        return -1;
+    }
+    long qc = 0;
+    if (cur != NULL) {
+        qc = PyLong_AsLong(cur);
+        if (qc == -1 && PyErr_Occurred()) {
+            Py_DECREF(counters);
+            return -1;
+        }
+    }
+    PyObject *nv = PyLong_FromLong(qc + 1);
+    if (nv == NULL ||
+        PyDict_SetItem(counters, s_queued_claim, nv) < 0) {
+        Py_XDECREF(nv);
+        Py_DECREF(counters);
+        return -1;
+    }
+    Py_DECREF(nv);
+    Py_DECREF(counters);
+
+    /* rebalance() unless one is already scheduled (the common case
+     * under sustained queueing) */
+    PyObject *sched = PyObject_GetAttr(t->ct_pool, s_p_rebal_scheduled);
+    if (sched == NULL)
+        return -1;
+    int pending = PyObject_IsTrue(sched);
+    Py_DECREF(sched);
+    if (pending < 0)
+        return -1;
+    if (!pending) {
+        PyObject *r = PyObject_CallMethodObjArgs(t->ct_pool,
+                                                 s_rebalance, NULL);
+        if (r == NULL)
+            return -1;
+        Py_DECREF(r);
+    }
+    return 0;
+}
+
 static int
 ct_slow(CTOb *t)
 {
@@ -3171,6 +3265,10 @@ ct_try_next(CTOb *t)
         Py_DECREF(r);
         return ct_note_demand(t);
     }
+    /* idle queue empty: queue as a waiter (C) unless errorOnEmpty
+     * semantics apply, which keep the full python logic */
+    if (!t->ct_err_on_empty)
+        return ct_enqueue_waiter(t);
     return ct_slow(t);
 }
 
@@ -4183,6 +4281,10 @@ PyInit__speed(void)
     s_p_busy_hwm = PyUnicode_InternFromString("p_busy_hwm");
     s_p_demand_hwm = PyUnicode_InternFromString("p_demand_hwm");
     s_ticket_slow = PyUnicode_InternFromString("_ticket_slow");
+    s_p_counters = PyUnicode_InternFromString("p_counters");
+    s_p_rebal_scheduled = PyUnicode_InternFromString("p_rebal_scheduled");
+    s_max_claim_queue = PyUnicode_InternFromString("max-claim-queue");
+    s_queued_claim = PyUnicode_InternFromString("queued-claim");
     s_connected_st = PyUnicode_InternFromString("connected");
     s_busy_st = PyUnicode_InternFromString("busy");
     s_stopping_st = PyUnicode_InternFromString("stopping");
