@@ -617,25 +617,7 @@ Scope_on(PyObject *self_, PyObject *const *args, Py_ssize_t nargs)
     }
     if (scope_check_active(self, "S.on()") < 0)
         return NULL;
-    PyObject *emitter = args[0], *event = args[1], *cb = args[2];
-    PyObject *r;
-    /* fast path for our own emitter type with no overridden .on */
-    if (Py_TYPE(emitter) == &EmitterType) {
-        r = emitter_add((Emitter *)emitter, event, cb);
-    } else {
-        r = PyObject_CallMethodObjArgs(emitter, s_on, event, cb, NULL);
-    }
-    if (r == NULL)
-        return NULL;
-    Py_DECREF(r);
-    if (self->sc_listeners == NULL) {
-        self->sc_listeners = PyList_New(0);
-        if (self->sc_listeners == NULL)
-            return NULL;
-    }
-    if (PyList_Append(self->sc_listeners, emitter) < 0 ||
-        PyList_Append(self->sc_listeners, event) < 0 ||
-        PyList_Append(self->sc_listeners, cb) < 0)
+    if (scope_on_c(self, args[0], args[1], args[2]) < 0)
         return NULL;
     Py_RETURN_NONE;
 }
@@ -3045,6 +3027,34 @@ typedef struct {
 
 extern PyTypeObject CTType;
 
+/* cache the pool's queues on the ticket; a pool whose queues are not
+ * the native deque leaves them NULL and the ticket permanently
+ * delegates to the python slow path */
+static void
+ct_bind_queues(CTOb *t, PyObject *pool)
+{
+    PyObject *iq = PyObject_GetAttrString(pool, "p_idleq");
+    PyObject *wq = iq ? PyObject_GetAttrString(pool, "p_waiters") : NULL;
+    PyObject *nq = wq ? PyObject_GetAttrString(pool, "p_initq") : NULL;
+    if (nq == NULL) {
+        PyErr_Clear();
+        Py_XDECREF(iq);
+        Py_XDECREF(wq);
+        return;
+    }
+    if (PyObject_TypeCheck(iq, &NQueueType) &&
+        PyObject_TypeCheck(wq, &NQueueType) &&
+        PyObject_TypeCheck(nq, &NQueueType)) {
+        t->ct_idleq = (NQueue *)iq;
+        t->ct_waiters = (NQueue *)wq;
+        t->ct_initq = (NQueue *)nq;
+    } else {
+        Py_DECREF(iq);
+        Py_DECREF(wq);
+        Py_DECREF(nq);
+    }
+}
+
 static int
 ct_note_demand(CTOb *t)
 {
@@ -3351,27 +3361,7 @@ CT_new(PyTypeObject *type, PyObject *args, PyObject *kwds)
     t->ct_initq = NULL;
     t->ct_err_on_empty = PyObject_IsTrue(err_on_empty);
     PyObject_GC_Track((PyObject *)t);
-
-    /* cache the pool's queues; if they are not the native deque the
-     * ticket permanently delegates to the python slow path */
-    PyObject *iq = PyObject_GetAttrString(pool, "p_idleq");
-    PyObject *wq = iq ? PyObject_GetAttrString(pool, "p_waiters") : NULL;
-    PyObject *nq = wq ? PyObject_GetAttrString(pool, "p_initq") : NULL;
-    if (nq == NULL) {
-        PyErr_Clear();
-        Py_XDECREF(iq);
-        Py_XDECREF(wq);
-    } else if (PyObject_TypeCheck(iq, &NQueueType) &&
-               PyObject_TypeCheck(wq, &NQueueType) &&
-               PyObject_TypeCheck(nq, &NQueueType)) {
-        t->ct_idleq = (NQueue *)iq;
-        t->ct_waiters = (NQueue *)wq;
-        t->ct_initq = (NQueue *)nq;
-    } else {
-        Py_DECREF(iq);
-        Py_DECREF(wq);
-        Py_DECREF(nq);
-    }
+    ct_bind_queues(t, pool);
     return (PyObject *)t;
 }
 
@@ -3441,24 +3431,7 @@ speed_claim_fast(PyObject *mod, PyObject *const *args, Py_ssize_t nargs)
     t->ct_initq = NULL;
     t->ct_err_on_empty = err_on_empty;
     PyObject_GC_Track((PyObject *)t);
-    PyObject *iq = PyObject_GetAttrString(pool, "p_idleq");
-    PyObject *wq = iq ? PyObject_GetAttrString(pool, "p_waiters") : NULL;
-    PyObject *nq = wq ? PyObject_GetAttrString(pool, "p_initq") : NULL;
-    if (nq == NULL) {
-        PyErr_Clear();
-        Py_XDECREF(iq);
-        Py_XDECREF(wq);
-    } else if (PyObject_TypeCheck(iq, &NQueueType) &&
-               PyObject_TypeCheck(wq, &NQueueType) &&
-               PyObject_TypeCheck(nq, &NQueueType)) {
-        t->ct_idleq = (NQueue *)iq;
-        t->ct_waiters = (NQueue *)wq;
-        t->ct_initq = (NQueue *)nq;
-    } else {
-        Py_DECREF(iq);
-        Py_DECREF(wq);
-        Py_DECREF(nq);
-    }
+    ct_bind_queues(t, pool);
     r = emitter_add((Emitter *)handle, s_stateChanged, (PyObject *)t);
     Py_DECREF((PyObject *)t);
     if (r == NULL) {
