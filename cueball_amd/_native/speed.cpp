@@ -879,11 +879,21 @@ fsm_get_loop_of(FSMOb *fsm)
     return fsm->f_loop;
 }
 
+int fsm_schedule_flush(FSMOb *self);
+
+/* Per-FSM drain cap: one FSM whose listeners keep triggering its own
+ * transitions would otherwise drain forever inside a single loop
+ * callback (the batch cap only bounds distinct flush entries).  The
+ * remainder is rescheduled in order. */
+#define FSM_FLUSH_CAP 64
+
 int
 fsm_flush_core(FSMOb *self)
 {
     self->f_emit_scheduled = 0;
-    while (PyList_GET_SIZE(self->f_emit_queue) > 0) {
+    int n = 0;
+    while (PyList_GET_SIZE(self->f_emit_queue) > 0 &&
+           n++ < FSM_FLUSH_CAP) {
         PyObject *st = PyList_GET_ITEM(self->f_emit_queue, 0);
         Py_INCREF(st);
         if (PyList_SetSlice(self->f_emit_queue, 0, 1, NULL) < 0) {
@@ -896,6 +906,9 @@ fsm_flush_core(FSMOb *self)
         if (r < 0)
             return -1;
     }
+    if (PyList_GET_SIZE(self->f_emit_queue) > 0 &&
+        !self->f_emit_scheduled)
+        return fsm_schedule_flush(self);
     return 0;
 }
 
@@ -1034,6 +1047,12 @@ fsm_queue_state_changed(FSMOb *self, PyObject *state)
 {
     if (PyList_Append(self->f_emit_queue, state) < 0)
         return -1;
+    return fsm_schedule_flush(self);
+}
+
+int
+fsm_schedule_flush(FSMOb *self)
+{
     if (self->f_emit_scheduled)
         return 0;
     self->f_emit_scheduled = 1;

@@ -337,12 +337,22 @@ class FSM(EventEmitter):
             self._fsm_emit_scheduled = True
             self._loop.call_soon(self._flush_state_changed)
 
+    #: per-flush drain cap: a listener that keeps re-triggering this
+    #: FSM's own transitions must not hog one loop callback forever
+    #: (mirrors the native core's FSM_FLUSH_CAP)
+    FLUSH_CAP = 64
+
     def _flush_state_changed(self) -> None:
         self._fsm_emit_scheduled = False
         q = self._fsm_emit_queue
-        while q:
+        n = 0
+        while q and n < self.FLUSH_CAP:
             st = q.pop(0)
             self.emit("stateChanged", st)
+            n += 1
+        if q and not self._fsm_emit_scheduled:
+            self._fsm_emit_scheduled = True
+            self._loop.call_soon(self._flush_state_changed)
 
 
 # ---------------------------------------------------------------------------

@@ -25,7 +25,7 @@ class PingPong(FSM):
 
 def test_chain_yields_to_other_callbacks():
     async def body(loop):
-        fsm = PingPong(loop, limit=5000)
+        fsm = PingPong(loop, limit=40000)
         ticks = []
 
         def on_change(st):
@@ -46,7 +46,7 @@ def test_chain_yields_to_other_callbacks():
 
         # drive until the chain finishes
         import asyncio
-        for _ in range(2000):
+        for _ in range(40000):
             if fsm.hops >= fsm.limit:
                 break
             await asyncio.sleep(0)
@@ -59,7 +59,7 @@ def test_chain_yields_to_other_callbacks():
 
 def test_chain_completes_in_bounded_loop_turns():
     async def body(loop):
-        fsm = PingPong(loop, limit=4000)
+        fsm = PingPong(loop, limit=20000)
 
         def on_change(st):
             fsm.hops += 1
@@ -70,7 +70,7 @@ def test_chain_completes_in_bounded_loop_turns():
 
         import asyncio
         turns = 0
-        while fsm.hops < fsm.limit and turns < 4000:
+        while fsm.hops < fsm.limit and turns < 20000:
             await asyncio.sleep(0)
             turns += 1
         assert fsm.hops >= fsm.limit
