@@ -69,7 +69,33 @@ function forEachPipeline(opts, cb) {
 	return (results);
 }
 
+function barrier() {
+	const EventEmitter = require('events').EventEmitter;
+	const b = new EventEmitter();
+	b.pending = {};
+	var count = 0;
+	b.start = function (name) {
+		if (!b.pending[name]) {
+			b.pending[name] = true;
+			count++;
+		}
+	};
+	b.done = function (name) {
+		if (b.pending[name]) {
+			delete b.pending[name];
+			if (--count === 0) {
+				setImmediate(function () {
+					if (count === 0)
+						b.emit('drain');
+				});
+			}
+		}
+	};
+	return (b);
+}
+
 module.exports = {
 	forEachParallel: forEachParallel,
-	forEachPipeline: forEachPipeline
+	forEachPipeline: forEachPipeline,
+	barrier: barrier
 };

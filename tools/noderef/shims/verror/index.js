@@ -61,6 +61,10 @@ VError.cause = function (err) {
 	return (err instanceof VError ? (err.jse_cause || null) : null);
 };
 
+VError.hasCauseWithName = function (err, name) {
+	return (VError.findCauseByName(err, name) !== null);
+};
+
 VError.findCauseByName = function (err, name) {
 	var e = err;
 	while (e) {
