@@ -21,6 +21,7 @@
 #define PY_SSIZE_T_CLEAN
 #include <Python.h>
 #include <structmember.h>
+#include <cmath>
 
 namespace {
 
@@ -3797,6 +3798,297 @@ slotkit_maybe_entry(PyObject *kit_, FSMOb *fsm, PyObject *target,
 }
 
 /* ------------------------------------------------------------------ */
+/* ControlledDelay: the CoDel AQM (codel.py / lib/codel.js) in C so    */
+/* the overload-shed feed path stays native.                           */
+
+#define CODEL_INTERVAL_MS 100.0
+
+typedef struct {
+    PyObject_HEAD
+    PyObject *cd_loop;          /* owned */
+    double cd_targdelay;
+    double cd_first_above_time;
+    double cd_drop_next;
+    long cd_count;
+    int cd_dropping;
+    double cd_last_empty;
+} CoDelOb;
+
+extern PyTypeObject CoDelType;
+
+static double
+codel_now(CoDelOb *self, int *err)
+{
+    PyObject *t = PyObject_CallMethodObjArgs(self->cd_loop, s_time, NULL);
+    if (t == NULL) {
+        *err = 1;
+        return 0.0;
+    }
+    double secs = PyFloat_AsDouble(t);
+    Py_DECREF(t);
+    if (secs == -1.0 && PyErr_Occurred()) {
+        *err = 1;
+        return 0.0;
+    }
+    *err = 0;
+    return secs * 1000.0;
+}
+
+static int
+codel_can_drop(CoDelOb *self, double now, double start)
+{
+    double sojourn = now - start;
+    if (sojourn < self->cd_targdelay)
+        self->cd_first_above_time = 0.0;
+    else if (self->cd_first_above_time == 0.0)
+        self->cd_first_above_time = now + CODEL_INTERVAL_MS;
+    else if (now >= self->cd_first_above_time)
+        return 1;
+    return 0;
+}
+
+/* int result: 1 drop, 0 keep, -1 error.  Mirrors
+ * ControlledDelay.overloaded() exactly. */
+static int
+codel_overloaded_c(CoDelOb *self, double start)
+{
+    int err = 0;
+    double now = codel_now(self, &err);
+    if (err)
+        return -1;
+    int ok_to_drop = codel_can_drop(self, now, start);
+    int drop_claim = 0;
+    if (self->cd_dropping) {
+        if (!ok_to_drop) {
+            self->cd_dropping = 0;
+        } else if (now >= self->cd_drop_next) {
+            /* NOTE: the reference deliberately does NOT advance
+             * cd_drop_next here (lib/codel.js:62-68): once dropping
+             * and past drop-next, every dequeue above target drops
+             * until the sojourn falls below target again. */
+            drop_claim = 1;
+            self->cd_count++;
+        }
+    } else if (ok_to_drop &&
+               ((now - self->cd_drop_next < CODEL_INTERVAL_MS) ||
+                (now - self->cd_first_above_time >=
+                 CODEL_INTERVAL_MS))) {
+        drop_claim = 1;
+        self->cd_dropping = 1;
+        if (now - self->cd_drop_next < CODEL_INTERVAL_MS)
+            self->cd_count = self->cd_count > 2 ?
+                self->cd_count - 2 : 1;
+        else
+            self->cd_count = 1;
+        self->cd_drop_next = now +
+            CODEL_INTERVAL_MS / sqrt((double)self->cd_count);
+    }
+    return drop_claim;
+}
+
+static PyObject *
+CoDel_overloaded(PyObject *self_, PyObject *start_)
+{
+    double start = PyFloat_AsDouble(start_);
+    if (start == -1.0 && PyErr_Occurred())
+        return NULL;
+    int r = codel_overloaded_c((CoDelOb *)self_, start);
+    if (r < 0)
+        return NULL;
+    return PyBool_FromLong(r);
+}
+
+static PyObject *
+CoDel_can_drop(PyObject *self_, PyObject *const *args, Py_ssize_t nargs)
+{
+    if (nargs != 2) {
+        PyErr_SetString(PyExc_TypeError, "can_drop(now, start)");
+        return NULL;
+    }
+    double now = PyFloat_AsDouble(args[0]);
+    if (now == -1.0 && PyErr_Occurred())
+        return NULL;
+    double start = PyFloat_AsDouble(args[1]);
+    if (start == -1.0 && PyErr_Occurred())
+        return NULL;
+    return PyBool_FromLong(codel_can_drop((CoDelOb *)self_, now, start));
+}
+
+static PyObject *
+CoDel_get_drop_next(PyObject *self_, PyObject *now_)
+{
+    CoDelOb *self = (CoDelOb *)self_;
+    double now = PyFloat_AsDouble(now_);
+    if (now == -1.0 && PyErr_Occurred())
+        return NULL;
+    return PyFloat_FromDouble(
+        now + CODEL_INTERVAL_MS / sqrt((double)self->cd_count));
+}
+
+static int
+codel_empty_c(CoDelOb *self)
+{
+    int err = 0;
+    double now = codel_now(self, &err);
+    if (err)
+        return -1;
+    self->cd_last_empty = now;
+    self->cd_first_above_time = 0.0;
+    return 0;
+}
+
+static PyObject *
+CoDel_empty(PyObject *self_, PyObject *noargs)
+{
+    (void)noargs;
+    if (codel_empty_c((CoDelOb *)self_) < 0)
+        return NULL;
+    Py_RETURN_NONE;
+}
+
+static PyObject *
+CoDel_get_max_idle(PyObject *self_, PyObject *noargs)
+{
+    CoDelOb *self = (CoDelOb *)self_;
+    (void)noargs;
+    double bound = self->cd_targdelay * 10.0;
+    int err = 0;
+    double now = codel_now(self, &err);
+    if (err)
+        return NULL;
+    if (self->cd_last_empty < now - bound)
+        return PyFloat_FromDouble(self->cd_targdelay * 3.0);
+    return PyFloat_FromDouble(bound);
+}
+
+static PyObject *
+CoDel_get_dropping(PyObject *self_, void *closure)
+{
+    (void)closure;
+    return PyBool_FromLong(((CoDelOb *)self_)->cd_dropping);
+}
+
+static int
+CoDel_traverse(PyObject *self_, visitproc visit, void *arg)
+{
+    Py_VISIT(((CoDelOb *)self_)->cd_loop);
+    return 0;
+}
+
+static int
+CoDel_clear_(PyObject *self_)
+{
+    Py_CLEAR(((CoDelOb *)self_)->cd_loop);
+    return 0;
+}
+
+static void
+CoDel_dealloc(PyObject *self_)
+{
+    PyObject_GC_UnTrack(self_);
+    CoDel_clear_(self_);
+    PyObject_GC_Del(self_);
+}
+
+static PyObject *
+CoDel_new(PyTypeObject *type, PyObject *args, PyObject *kwds)
+{
+    PyObject *target, *loop = Py_None;
+    static const char *kwlist[] = {"target_claim_delay", "loop", NULL};
+    if (!PyArg_ParseTupleAndKeywords(args, kwds, "O|O",
+                                     const_cast<char **>(kwlist),
+                                     &target, &loop))
+        return NULL;
+    double targ;
+    if (PyFloat_Check(target) || PyLong_Check(target)) {
+        targ = PyFloat_AsDouble(target);
+        if (targ == -1.0 && PyErr_Occurred())
+            return NULL;
+    } else {
+        PyErr_SetString(PyExc_ValueError,
+                        "target_claim_delay must be finite");
+        return NULL;
+    }
+    if (!std::isfinite(targ)) {
+        PyErr_SetString(PyExc_ValueError,
+                        "target_claim_delay must be finite");
+        return NULL;
+    }
+    PyObject *resolved = PyObject_CallOneArg(g_get_loop, loop);
+    if (resolved == NULL)
+        return NULL;
+    CoDelOb *self = PyObject_GC_New(CoDelOb, type);
+    if (self == NULL) {
+        Py_DECREF(resolved);
+        return NULL;
+    }
+    self->cd_loop = resolved;
+    self->cd_targdelay = targ;
+    self->cd_first_above_time = 0.0;
+    self->cd_drop_next = 0.0;
+    self->cd_count = 0;
+    self->cd_dropping = 0;
+    self->cd_last_empty = 0.0;
+    PyObject_GC_Track((PyObject *)self);
+    /* treat the queue as having just been empty (see codel.py) */
+    int err = 0;
+    double now = codel_now(self, &err);
+    if (err) {
+        Py_DECREF((PyObject *)self);
+        return NULL;
+    }
+    self->cd_last_empty = now;
+    return (PyObject *)self;
+}
+
+static PyMethodDef CoDel_methods[] = {
+    {"overloaded", CoDel_overloaded, METH_O, NULL},
+    {"can_drop", (PyCFunction)(void (*)(void))CoDel_can_drop,
+     METH_FASTCALL, NULL},
+    {"get_drop_next", CoDel_get_drop_next, METH_O, NULL},
+    {"empty", CoDel_empty, METH_NOARGS, NULL},
+    {"get_max_idle", CoDel_get_max_idle, METH_NOARGS, NULL},
+    {NULL, NULL, 0, NULL},
+};
+
+static PyMemberDef CoDel_members[] = {
+    {(char *)"cd_targdelay", T_DOUBLE, offsetof(CoDelOb, cd_targdelay),
+     READONLY, NULL},
+    {(char *)"cd_first_above_time", T_DOUBLE,
+     offsetof(CoDelOb, cd_first_above_time), 0, NULL},
+    {(char *)"cd_drop_next", T_DOUBLE, offsetof(CoDelOb, cd_drop_next),
+     0, NULL},
+    {(char *)"cd_count", T_LONG, offsetof(CoDelOb, cd_count), 0, NULL},
+    {(char *)"cd_last_empty", T_DOUBLE,
+     offsetof(CoDelOb, cd_last_empty), 0, NULL},
+    {NULL, 0, 0, 0, NULL},
+};
+
+static PyGetSetDef CoDel_getset[] = {
+    {(char *)"cd_dropping", CoDel_get_dropping, NULL, NULL, NULL},
+    {NULL, NULL, NULL, NULL, NULL},
+};
+
+PyTypeObject CoDelType = {
+    PyVarObject_HEAD_INIT(NULL, 0)
+    "cueball_amd._speed.ControlledDelay",
+    sizeof(CoDelOb),
+};
+
+static void
+codel_type_init(void)
+{
+    CoDelType.tp_dealloc = CoDel_dealloc;
+    CoDelType.tp_flags = Py_TPFLAGS_DEFAULT | Py_TPFLAGS_HAVE_GC;
+    CoDelType.tp_traverse = CoDel_traverse;
+    CoDelType.tp_clear = CoDel_clear_;
+    CoDelType.tp_methods = CoDel_methods;
+    CoDelType.tp_members = CoDel_members;
+    CoDelType.tp_getset = CoDel_getset;
+    CoDelType.tp_new = CoDel_new;
+}
+
+/* ------------------------------------------------------------------ */
 /* SlotDispatch: the pool's per-slot stateChanged listener             */
 /* (pool._slot_state_changed) with the idle-feed and busy no-op fast   */
 /* paths in C; every other state (and every unusual condition)         */
@@ -3818,7 +4110,9 @@ typedef struct {
     PyObject *sd_key;       /* owned */
     NQueue *sd_idleq;       /* owned (never replaced by the pool) */
     NQueue *sd_waiters;     /* owned */
-    int sd_has_codel;       /* CoDel pools use the python path */
+    PyObject *sd_codel;     /* owned native ControlledDelay or NULL */
+    int sd_has_codel;       /* codel configured but not native =>
+                             * python path */
 } SlotDispatch;
 
 extern PyTypeObject SlotDispatchType;
@@ -3882,7 +4176,7 @@ SlotDispatch_call(PyObject *self_, PyObject *args, PyObject *kwds)
         Py_RETURN_NONE;
     }
 
-    if (!in_initq && !d->sd_has_codel &&
+    if (!in_initq && (!d->sd_has_codel || d->sd_codel != NULL) &&
         (st == s_idle ||
          (PyUnicode_Check(st) &&
           PyUnicode_Compare(st, s_idle) == 0))) {
@@ -3960,14 +4254,36 @@ SlotDispatch_call(PyObject *self_, PyObject *args, PyObject *kwds)
                 Py_RETURN_NONE;
             }
         }
-        /* feed waiters */
+        /* feed waiters (with the CoDel drop check on each; like the
+         * python dispatcher, overloaded() is fed BEFORE the staleness
+         * check so the controller sees every dequeue) */
         while (d->sd_waiters->q_len > 0) {
             PyObject *hdl = nqueue_shift_value(d->sd_waiters);
             if (hdl == NULL)
                 return NULL;
-            if (!PyObject_TypeCheck(hdl, &CHType) ||
-                !ch_state_is((CHOb *)hdl, s_waiting)) {
+            if (!PyObject_TypeCheck(hdl, &CHType)) {
                 Py_DECREF(hdl);
+                continue;
+            }
+            int drop = 0;
+            if (d->sd_codel != NULL) {
+                drop = codel_overloaded_c((CoDelOb *)d->sd_codel,
+                                          ((CHOb *)hdl)->chb_started);
+                if (drop < 0) {
+                    Py_DECREF(hdl);
+                    return NULL;
+                }
+            }
+            if (!ch_state_is((CHOb *)hdl, s_waiting)) {
+                Py_DECREF(hdl);
+                continue;
+            }
+            if (drop) {
+                PyObject *r = CH_timeout(hdl, NULL);
+                Py_DECREF(hdl);
+                if (r == NULL)
+                    return NULL;
+                Py_DECREF(r);
                 continue;
             }
             PyObject *r = CH_try_(hdl, d->sd_fsm);
@@ -3977,6 +4293,9 @@ SlotDispatch_call(PyObject *self_, PyObject *args, PyObject *kwds)
             Py_DECREF(r);
             Py_RETURN_NONE;
         }
+        if (d->sd_codel != NULL &&
+            codel_empty_c((CoDelOb *)d->sd_codel) < 0)
+            return NULL;
         /* no waiter: park on the idle queue */
         QNode *n = nqueue_push(d->sd_idleq, d->sd_fsm);
         if (n == NULL)
@@ -4003,6 +4322,7 @@ SlotDispatch_traverse(PyObject *self_, visitproc visit, void *arg)
     Py_VISIT(d->sd_key);
     Py_VISIT((PyObject *)d->sd_idleq);
     Py_VISIT((PyObject *)d->sd_waiters);
+    Py_VISIT(d->sd_codel);
     return 0;
 }
 
@@ -4015,6 +4335,7 @@ SlotDispatch_clear_(PyObject *self_)
     Py_CLEAR(d->sd_key);
     Py_CLEAR(d->sd_idleq);
     Py_CLEAR(d->sd_waiters);
+    Py_CLEAR(d->sd_codel);
     return 0;
 }
 
@@ -4065,7 +4386,18 @@ SlotDispatch_new(PyTypeObject *type, PyObject *args, PyObject *kwds)
     d->sd_idleq = (NQueue *)iq;
     d->sd_waiters = (NQueue *)wq;
     d->sd_has_codel = PyObject_IsTrue(has_codel);
+    d->sd_codel = NULL;
     PyObject_GC_Track((PyObject *)d);
+    if (d->sd_has_codel) {
+        PyObject *cd = PyObject_GetAttrString(pool, "p_codel");
+        if (cd == NULL) {
+            PyErr_Clear();
+        } else if (PyObject_TypeCheck(cd, &CoDelType)) {
+            d->sd_codel = cd;     /* native: feed path stays in C */
+        } else {
+            Py_DECREF(cd);        /* pure codel: python feed path */
+        }
+    }
     return (PyObject *)d;
 }
 
@@ -4308,6 +4640,7 @@ PyInit__speed(void)
     queue_types_init();
     ct_type_init();
     slotkit_types_init();
+    codel_type_init();
     if (PyType_Ready(&EmitterType) < 0 ||
         PyType_Ready(&OnceWrapperType) < 0 ||
         PyType_Ready(&GuardedCbType) < 0 ||
@@ -4323,7 +4656,8 @@ PyInit__speed(void)
         PyType_Ready(&CTType) < 0 ||
         PyType_Ready(&KitCbType) < 0 ||
         PyType_Ready(&SlotKitType) < 0 ||
-        PyType_Ready(&SlotDispatchType) < 0)
+        PyType_Ready(&SlotDispatchType) < 0 ||
+        PyType_Ready(&CoDelType) < 0)
         return NULL;
 
     g_remove_desc = PyDict_GetItemString(EmitterType.tp_dict,
@@ -4353,5 +4687,7 @@ PyInit__speed(void)
     PyModule_AddObject(m, "SlotKit", (PyObject *)&SlotKitType);
     Py_INCREF(&SlotDispatchType);
     PyModule_AddObject(m, "SlotDispatch", (PyObject *)&SlotDispatchType);
+    Py_INCREF(&CoDelType);
+    PyModule_AddObject(m, "ControlledDelay", (PyObject *)&CoDelType);
     return m;
 }

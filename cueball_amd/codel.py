@@ -95,3 +95,19 @@ class ControlledDelay:
         if self.cd_last_empty < now - bound:
             return self.cd_targdelay * 3.0
         return bound
+
+
+# ---------------------------------------------------------------------------
+# Native core: the C ControlledDelay replaces the pure-Python one
+# above when the extension is built — identical algorithm (the CoDel
+# suites run against either; CUEBALL_PURE=1 forces the Python class,
+# which also lets the pool's python feed path exercise CoDel).
+import os as _os
+
+PurePythonControlledDelay = ControlledDelay
+
+if not _os.environ.get("CUEBALL_PURE"):
+    try:
+        from ._speed import ControlledDelay  # type: ignore # noqa: F811
+    except ImportError:
+        pass
