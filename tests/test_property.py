@@ -51,7 +51,7 @@ def test_pool_survives_random_interleavings(actions):
         loop.close()
 
 
-async def _scenario(loop, actions, with_checker=False):
+async def _scenario(loop, actions, with_checker=False, with_codel=False):
     conns = []
     resolver = DummyResolver()
     rfsm = ResolverFSM(resolver, {"loop": loop})
@@ -88,6 +88,12 @@ async def _scenario(loop, actions, with_checker=False):
 
         pool_opts["checkTimeout"] = 150
         pool_opts["checker"] = checker
+    if with_codel:
+        # exercises the CoDel feed path (native when built): claims
+        # get the adaptive max-idle timeout; explicit timeouts are
+        # forbidden by the API, so the 'claim' action degrades to the
+        # no-options shape below
+        pool_opts["targetClaimDelay"] = 20
     pool = ConnectionPool(pool_opts)
     rfsm.start()
 
@@ -137,7 +143,8 @@ async def _scenario(loop, actions, with_checker=False):
                 live[seed % len(live)].emit("close")
         elif kind == "claim":
             box = cb_box()
-            box["handle_obj"] = pool.claim({"timeout": 400}, box["cb"])
+            opts = {} if with_codel else {"timeout": 400}
+            box["handle_obj"] = pool.claim(opts, box["cb"])
             pending.append(box)
         elif kind == "claim_inf":
             # no timeout: resolution guaranteed only by feed, failure,
@@ -378,5 +385,21 @@ def test_pool_with_checker_survives_interleavings(actions):
     try:
         loop.run_until_complete(
             _scenario(loop, actions, with_checker=True))
+    finally:
+        loop.close()
+
+
+@settings(max_examples=120, deadline=None,
+          suppress_health_check=[HealthCheck.too_slow])
+@given(actions=ACTIONS)
+def test_pool_with_codel_survives_interleavings(actions):
+    """Same interleavings against a targetClaimDelay pool: the CoDel
+    controller (native when built) sees every dequeue, drops surface
+    as claim timeouts, and the end-of-run invariants (every callback
+    fired exactly once, pool stops clean) still hold."""
+    loop = VirtualLoop()
+    try:
+        loop.run_until_complete(
+            _scenario(loop, actions, with_codel=True))
     finally:
         loop.close()
