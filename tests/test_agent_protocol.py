@@ -2,8 +2,6 @@
 (lib/agent.js:296-396): agentRemove+abort and abort-after-free
 interleavings, driven deterministically with scripted fakes."""
 
-import asyncio
-
 from cueball_amd.agent import _RequestTicket
 from cueball_amd.events import EventEmitter
 from conftest import run_vt

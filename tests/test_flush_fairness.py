@@ -26,7 +26,6 @@ class PingPong(FSM):
 def test_chain_yields_to_other_callbacks():
     async def body(loop):
         fsm = PingPong(loop, limit=40000)
-        ticks = []
 
         def on_change(st):
             fsm.hops += 1
