@@ -3391,6 +3391,86 @@ PyTypeObject CTType = {
     sizeof(CTOb),
 };
 
+/* pool_claim(handle_cls, pool, cb, log, loop) -> handle | True
+ * The whole no-options/no-CoDel pool.claim() body in C: bump the
+ * claim counter, check the pool state (returns True when
+ * stopping/stopped/failed so python builds the short-circuit error
+ * without re-counting), capture the claim stack, and build
+ * handle+ticket via the claim_fast core with timeout=inf. */
+static PyObject *speed_claim_fast(PyObject *mod, PyObject *const *args,
+                                  Py_ssize_t nargs);
+
+static PyObject *
+speed_pool_claim(PyObject *mod, PyObject *const *args, Py_ssize_t nargs)
+{
+    (void)mod;
+    if (nargs != 5) {
+        PyErr_SetString(PyExc_TypeError,
+                        "pool_claim(cls, pool, cb, log, loop)");
+        return NULL;
+    }
+    PyObject *pool = args[1];
+    if (!PyObject_TypeCheck(pool, &FSMType)) {
+        PyErr_SetString(PyExc_TypeError, "pool must be a native FSM");
+        return NULL;
+    }
+    /* counters["claim"] += 1 ("claim" is not a tracked metric event) */
+    PyObject *counters = PyObject_GetAttr(pool, s_p_counters);
+    if (counters == NULL)
+        return NULL;
+    if (PyDict_Check(counters)) {
+        PyObject *cur = PyDict_GetItemWithError(counters, s_claim);
+        if (cur == NULL && PyErr_Occurred()) {
+            Py_DECREF(counters);
+            return NULL;
+        }
+        long v = 0;
+        if (cur != NULL) {
+            v = PyLong_AsLong(cur);
+            if (v == -1 && PyErr_Occurred()) {
+                Py_DECREF(counters);
+                return NULL;
+            }
+        }
+        PyObject *nv = PyLong_FromLong(v + 1);
+        if (nv == NULL ||
+            PyDict_SetItem(counters, s_claim, nv) < 0) {
+            Py_XDECREF(nv);
+            Py_DECREF(counters);
+            return NULL;
+        }
+        Py_DECREF(nv);
+    }
+    Py_DECREF(counters);
+
+    PyObject *pst = ((FSMOb *)pool)->f_state;
+    int bad = 0;
+    if (pst == NULL) {
+        bad = 1;
+    } else if (!(pst == s_running_st ||
+                 PyUnicode_Compare(pst, s_running_st) == 0)) {
+        /* only starting is also claim-able */
+        bad = !(PyUnicode_CompareWithASCIIString(pst, "starting") == 0);
+    }
+    if (bad)
+        Py_RETURN_TRUE;
+
+    PyObject *stack = PyObject_CallNoArgs(g_capture_stack);
+    if (stack == NULL)
+        return NULL;
+    PyObject *inf = PyFloat_FromDouble(HUGE_VAL);
+    if (inf == NULL) {
+        Py_DECREF(stack);
+        return NULL;
+    }
+    PyObject *cfargs[8] = {args[0], pool, stack, args[2], args[3],
+                           inf, args[4], Py_False};
+    PyObject *h = speed_claim_fast(NULL, cfargs, 8);
+    Py_DECREF(stack);
+    Py_DECREF(inf);
+    return h;
+}
+
 /* claim_fast(handle_cls, pool, stack, cb, log, timeout, loop,
  *            err_on_empty) -> handle
  * One C call for pool.claim's hot tail: allocate the ClaimHandle
@@ -4539,6 +4619,8 @@ PyMethodDef speed_methods[] = {
     {"_set_helpers", (PyCFunction)(void (*)(void))speed_set_helpers,
      METH_FASTCALL, NULL},
     {"claim_fast", (PyCFunction)(void (*)(void))speed_claim_fast,
+     METH_FASTCALL, NULL},
+    {"pool_claim", (PyCFunction)(void (*)(void))speed_pool_claim,
      METH_FASTCALL, NULL},
     {NULL, NULL, 0, NULL},
 };

@@ -106,11 +106,13 @@ class _ClaimTicket:
 
 _NativeClaimTicket = None
 _native_claim_fast = None
+_native_pool_claim = None
 _NativeSlotDispatch = None
 if not _os.environ.get("CUEBALL_PURE"):
     try:
         from ._speed import ClaimTicket as _NativeClaimTicket  # noqa: F811
         from ._speed import claim_fast as _native_claim_fast
+        from ._speed import pool_claim as _native_pool_claim
         from ._speed import SlotDispatch as _NativeSlotDispatch
     except ImportError:
         pass
@@ -869,6 +871,20 @@ class ConnectionPool(FSM):
         # fast path: claim({}, cb) / claim(cb) with no CoDel — the
         # overwhelmingly common call shape on the hot path
         if not options and cb is not None and self.p_codel is None:
+            if _native_pool_claim is not None:
+                # counter + state check + stack + handle/ticket in C;
+                # True means stopping/stopped/failed (counter already
+                # bumped): build the short-circuit error here
+                h = _native_pool_claim(ClaimHandle, self, cb,
+                                       self.p_claim_log, self._loop)
+                if h is not True:
+                    return h
+                if self._fsm_state == "failed":
+                    return self._claim_shortcircuit(
+                        cb, mod_errors.PoolFailedError(
+                            self, self.p_last_error))
+                return self._claim_shortcircuit(
+                    cb, mod_errors.PoolStoppingError(self))
             err_on_empty = False
             timeout = math.inf
         else:
