@@ -552,6 +552,30 @@ def reduce_results(results, rank, world):
     return out
 
 
+# The reference publishes no numbers, so the baseline is
+# node-cueball itself measured head-to-head on the MI355X benchmark box
+# via tools/noderef (profiles/headtohead_mi355x_v4.json).  Keyed by
+# (config, workers); vs_baseline is null for unmeasured combinations.
+_REFERENCE_MEASURED = {
+    ("headline", 1): 140955.0,
+    ("headline", 2): 281620.0,
+    ("headline", 4): 562049.0,
+    ("headline", 8): 1111609.0,
+    ("static1", 1): 138214.0,
+    ("dns", 1): 140377.0,
+    ("codel", 1): 47039.0,
+    ("agent", 1): 22107.0,
+    ("cset", 1): 221131.0,
+}
+
+
+def _vs_baseline(config, world, value):
+    ref = _REFERENCE_MEASURED.get((config, world))
+    if ref is None:
+        return None
+    return round(value / ref, 3)
+
+
 def main():
     ap = argparse.ArgumentParser()
     ap.add_argument("--gpus", type=int, default=1,
@@ -590,7 +614,7 @@ def main():
             "ms_per_step": round(ms_per_step, 3),
             "higher_is_better": True,
             "scaling": "weak",
-            "vs_baseline": None,
+            "vs_baseline": _vs_baseline(args.config, world, value),
             "dtype": "n/a",
             "data": "synthetic",
             "config": {
