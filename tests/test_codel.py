@@ -91,3 +91,39 @@ def test_get_max_idle_bounds():
         assert cd.get_max_idle() == 300
 
     run_vt(lambda loop: body(loop))
+
+
+def test_native_and_pure_controllers_agree():
+    """Drop-for-drop agreement between the native and pure CoDel
+    implementations over a scripted overload pattern (this caught the
+    drop-next-not-advanced reference quirk during the C port)."""
+    from cueball_amd.codel import (ControlledDelay,
+                                   PurePythonControlledDelay)
+    from cueball_amd.testing import VirtualLoop, advance
+
+    async def run(loop, cls):
+        cd = cls(5.0, loop=loop)
+        pattern = []
+        # overload, recover, overload again
+        for i in range(150):
+            sojourn = 50.0 if (i < 60 or i >= 100) else 1.0
+            start = loop.time() * 1000.0 - sojourn
+            pattern.append(bool(cd.overloaded(start)))
+            if i == 80:
+                cd.empty()
+            await advance(loop, 0.01)
+        return pattern, cd.cd_count, cd.cd_dropping
+
+    loop = VirtualLoop()
+    try:
+        nat = loop.run_until_complete(run(loop, ControlledDelay))
+    finally:
+        loop.close()
+    loop = VirtualLoop()
+    try:
+        pure = loop.run_until_complete(
+            run(loop, PurePythonControlledDelay))
+    finally:
+        loop.close()
+    assert nat == pure
+    assert any(nat[0]), "pattern should include drops"
