@@ -5,12 +5,16 @@
 # depends on — which is what makes the head-to-head benchmark a fair
 # comparison.
 #
-# Covered: utils, pool, cset, codel, resolver_for, resolver_static
-# (510 assertions).  Skipped: dns.test.js (needs the `mname` DNS
-# *server* package to synthesize wire responses), agent/monitor
-# (need restify/kang/sshpk HTTP stacks).  Those code paths are
-# exercised on the benchmark side by bench_ref.js's dns/agent
-# scenarios instead.
+# Covered strictly: utils, pool, cset, codel, resolver_for,
+# resolver_static (510 assertions).  dns.test.js also runs (61/62
+# assertions green) but is reported rather than enforced: its
+# "duped records" case asserts an exact number of TTL re-queries
+# inside fixed 1500 ms windows, and with the shims the resolver's
+# initial resolution completes slightly faster than with the real
+# npm stack, shifting the window phase (the re-query cadence itself
+# is correct: ~1.0-1.15 s for TTL=1 s plus the forward spread).
+# Skipped: agent/monitor (need restify/kang/sshpk HTTP stacks) —
+# those paths are exercised by bench_ref.js's agent scenario.
 #
 # Usage: bash tools/noderef/run_ref_tests.sh [reference-dir]
 
@@ -40,6 +44,8 @@ for f in utils codel resolver_for resolver_static pool cset; do
         fails=$((fails+1))
     fi
 done
+echo "== dns.test.js (informational; see header) =="
+(cd "$STAGE" && timeout 600 node test/dns.test.js | tail -4) || true
 rm -rf "$STAGE"
 if [ "$fails" -ne 0 ]; then
     echo "REFERENCE SUITE FAILURES: $fails file(s)"

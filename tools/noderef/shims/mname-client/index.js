@@ -261,4 +261,54 @@ DnsClient.prototype.lookup = function (opts, cb) {
 	maybeNext();
 };
 
-module.exports = { DnsClient: DnsClient };
+/* DnsMessage(parsedPacket): wrap an already-parsed packet object (the
+ * shape mname's Protocol produces; the reference's dns tests build
+ * these by hand) in the message interface the resolver consumes. */
+
+const TYPE_NAMES = QTYPE_REV;
+
+function convertRecord(rr) {
+	const out = { name: rr.name,
+	    type: TYPE_NAMES[rr.rtype] || rr.rtype,
+	    ttl: rr.rttl };
+	const rd = rr.rdata || {};
+	if (out.type === 'SRV') {
+		out.priority = rd.priority;
+		out.weight = rd.weight;
+		out.port = rd.port;
+		out.target = rd.target;
+	} else if (out.type === 'A' || out.type === 'AAAA' ||
+	    out.type === 'CNAME' || out.type === 'DNAME') {
+		out.target = rd.target;
+	} else if (out.type === 'SOA') {
+		out.ttl = rr.rttl;
+	}
+	return (out);
+}
+
+function DnsMessage(parsed) {
+	this.id = (parsed.header && parsed.header.id) || 0;
+	const flags = (parsed.header && parsed.header.flags) || {};
+	const rc = flags.rcode === undefined ? 0 : flags.rcode;
+	this.rcode = RCODES[rc] || ('RCODE' + rc);
+	this.m_answers = (parsed.answer || []).map(convertRecord);
+	this.m_authority = (parsed.authority || []).map(convertRecord);
+	this.m_additionals = (parsed.additional || []).map(convertRecord);
+}
+
+DnsMessage.prototype.getAnswers = function () {
+	return (this.m_answers);
+};
+DnsMessage.prototype.getAuthority = function () {
+	return (this.m_authority);
+};
+DnsMessage.prototype.getAdditionals = function () {
+	return (this.m_additionals);
+};
+DnsMessage.prototype.toError = function () {
+	if (this.rcode === 'NOERROR')
+		return (null);
+	return (new RcodeError('(parsed)', '(message)', this.rcode));
+};
+
+module.exports = { DnsClient: DnsClient, DnsMessage: DnsMessage };
