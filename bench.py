@@ -557,15 +557,15 @@ def reduce_results(results, rank, world):
 # via tools/noderef (profiles/headtohead_mi355x_v4.json).  Keyed by
 # (config, workers); vs_baseline is null for unmeasured combinations.
 _REFERENCE_MEASURED = {
-    ("headline", 1): 140955.0,
-    ("headline", 2): 281620.0,
-    ("headline", 4): 562049.0,
-    ("headline", 8): 1111609.0,
-    ("static1", 1): 138214.0,
-    ("dns", 1): 140377.0,
-    ("codel", 1): 47039.0,
-    ("agent", 1): 22107.0,
-    ("cset", 1): 221131.0,
+    ("headline", 1): 139403.0,
+    ("headline", 2): 291313.0,
+    ("headline", 4): 578116.0,
+    ("headline", 8): 1132238.0,
+    ("static1", 1): 140011.0,
+    ("dns", 1): 137741.0,
+    ("codel", 1): 50124.0,
+    ("agent", 1): 21704.0,
+    ("cset", 1): 250929.0,
 }
 
 
