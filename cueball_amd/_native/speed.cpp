@@ -934,7 +934,7 @@ FSM_flush_state_changed(PyObject *self_, PyObject *noargs)
  * turn (epoll_wait + ready-queue bookkeeping) per link.  The drain is
  * capped per callback so a self-sustaining chain cannot starve IO; the
  * remainder is rescheduled with a fresh call_soon. */
-#define FLUSH_DRAIN_CAP 256
+#define FLUSH_DRAIN_CAP 48
 
 typedef struct {
     PyObject_HEAD
