@@ -557,15 +557,15 @@ def reduce_results(results, rank, world):
 # via tools/noderef (profiles/headtohead_mi355x_v4.json).  Keyed by
 # (config, workers); vs_baseline is null for unmeasured combinations.
 _REFERENCE_MEASURED = {
-    ("headline", 1): 139403.0,
-    ("headline", 2): 291313.0,
-    ("headline", 4): 578116.0,
-    ("headline", 8): 1132238.0,
-    ("static1", 1): 140011.0,
-    ("dns", 1): 137741.0,
-    ("codel", 1): 50124.0,
-    ("agent", 1): 21704.0,
-    ("cset", 1): 250929.0,
+    ("headline", 1): 137252.0,
+    ("headline", 2): 282753.0,
+    ("headline", 4): 546788.0,
+    ("headline", 8): 1058642.0,
+    ("static1", 1): 135854.0,
+    ("dns", 1): 142658.0,
+    ("codel", 1): 54712.0,
+    ("agent", 1): 23478.0,
+    ("cset", 1): 225797.0,
 }
 
 
