@@ -1,0 +1,97 @@
+"""The claim path must produce ZERO cyclic garbage (native runtime).
+
+CPython's cyclic collector was measured to dominate claim p99 (4x)
+before the terminal-state cycle breaking; this pins the property so it
+cannot regress.  Runs on a real event loop with real sockets, like the
+benchmark."""
+
+import asyncio
+import gc
+
+import pytest
+
+from cueball_amd.events import NATIVE
+
+pytestmark = pytest.mark.skipif(not NATIVE, reason="native-only property")
+
+
+def test_claim_cycle_leaves_no_cyclic_garbage():
+    import bench
+
+    async def main():
+        backends = await bench.start_backends(4)
+        loop = asyncio.get_running_loop()
+        pool = bench.make_pool(backends, spares=4, maximum=8, loop=loop)
+        await bench.wait_for_idle(pool, 4)
+        lat = []
+        drv = bench.ClaimDriver(pool, loop, 8, lat)
+        await drv.run_step(2000)   # warm (pool growth allocates)
+
+        gc.collect()
+        gc.disable()
+        try:
+            await drv.run_step(5000)
+            collected = gc.collect()
+        finally:
+            gc.enable()
+        assert collected == 0, (
+            "%d cyclic objects left by 5000 claims" % collected)
+
+        pool.stop()
+        for s, _ in backends:
+            s.close()
+        await asyncio.sleep(0.1)
+
+    loop = asyncio.new_event_loop()
+    try:
+        loop.run_until_complete(main())
+    finally:
+        loop.close()
+
+
+def test_timed_out_claims_leave_no_cyclic_garbage():
+    """The shed/timeout path (codel-style) ends in the 'failed'
+    terminal state; it must be cycle-free too."""
+    from cueball_amd.pool import ConnectionPool
+    from cueball_amd.resolver import ResolverFSM
+    from cueball_amd.testing import DummyResolver, settle
+    from conftest import run_vt
+
+    async def body(loop):
+        resolver = DummyResolver()
+        rfsm = ResolverFSM(resolver, {"loop": loop})
+        pool = ConnectionPool({
+            "domain": "gc.test",
+            "constructor": lambda b: (_ for _ in ()).throw(
+                RuntimeError("never called")),
+            "recovery": {"default": {"timeout": 500, "retries": 1,
+                                     "delay": 0}},
+            "spares": 1,
+            "maximum": 1,
+            "resolver": rfsm,
+            "loop": loop,
+        })
+        rfsm.start()
+        await settle(loop)
+
+        fired = []
+        gc.collect()
+        gc.disable()
+        try:
+            for i in range(500):
+                pool.claim({"timeout": 5},
+                           lambda e, h=None, c=None: fired.append(e))
+            from cueball_amd.testing import advance
+            await advance(loop, 0.1)
+            collected = gc.collect()
+        finally:
+            gc.enable()
+        assert len(fired) == 500
+        assert all(e is not None for e in fired)
+        assert collected == 0, (
+            "%d cyclic objects left by 500 timed-out claims"
+            % collected)
+        pool.stop()
+        await settle(loop)
+
+    run_vt(lambda loop: body(loop))
