@@ -528,7 +528,34 @@ class ClaimHandle(FSM):
 
     def state_failed(self, S: StateScope) -> None:
         S.valid_transitions([])
-        S.immediate(lambda: self.ch_callback(self.ch_last_error))
+        # scheduled directly, not through the scope: failed is terminal
+        # so the callback can never become stale, and an empty terminal
+        # scope lets _fsm_terminal_settled dispose it (cycle break;
+        # mirrors the native CH_state_failed)
+        self._loop.call_soon(
+            lambda: self.ch_callback(self.ch_last_error))
+
+    def _fsm_terminal_settled(self) -> None:
+        """Break the per-claim reference cycles once the handle settles
+        in a terminal state (pure twin of ch_terminal_cleanup in
+        speed.cpp): dispose the empty terminal scope (fsm<->scope) and
+        unregister the pool ticket (handle<->ticket).  User listeners
+        are untouched and still receive the queued terminal
+        stateChanged."""
+        sc = self._fsm_scope
+        if sc is not None:
+            self._fsm_scope = None
+            sc._dispose()
+        ev = getattr(self, "_events", None)
+        if ev is None:
+            return
+        ls = ev.get("stateChanged")
+        if not ls:
+            return
+        ls[:] = [h for h in ls
+                 if not getattr(h, "_cueball_ticket", False)]
+        if not ls:
+            ev.pop("stateChanged", None)
 
 
 # the handle's own claimed-state error listener is not a user leak

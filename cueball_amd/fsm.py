@@ -329,6 +329,14 @@ class FSM(EventEmitter):
                         % (type(self).__name__, target, pend, valid)
                     )
                 next_state = pend
+        # settled in a terminal state (declared by an empty
+        # valid-transitions list): subclasses may break reference
+        # cycles here (ClaimHandle does — see the native twin's
+        # ch_terminal_cleanup)
+        if self._fsm_valid is not None and len(self._fsm_valid) == 0:
+            hook = getattr(self, "_fsm_terminal_settled", None)
+            if hook is not None:
+                hook()
 
     # -- async stateChanged delivery ------------------------------------
     def _queue_state_changed(self, state: str) -> None:

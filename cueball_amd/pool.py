@@ -93,6 +93,10 @@ class _ClaimTicket:
 
     __slots__ = ("pool", "handle", "err_on_empty")
 
+    #: lets the handle's terminal cleanup unregister us (see
+    #: ClaimHandle._fsm_terminal_settled)
+    _cueball_ticket = True
+
     def __init__(self, pool: "ConnectionPool", handle: ClaimHandle,
                  err_on_empty: bool) -> None:
         self.pool = pool

@@ -1,4 +1,4 @@
-"""The claim path must produce ZERO cyclic garbage (native runtime).
+"""The claim path must produce ZERO cyclic garbage (both runtimes).
 
 CPython's cyclic collector was measured to dominate claim p99 (4x)
 before the terminal-state cycle breaking; this pins the property so it
@@ -7,12 +7,6 @@ benchmark."""
 
 import asyncio
 import gc
-
-import pytest
-
-from cueball_amd.events import NATIVE
-
-pytestmark = pytest.mark.skipif(not NATIVE, reason="native-only property")
 
 
 def test_claim_cycle_leaves_no_cyclic_garbage():
