@@ -3155,6 +3155,7 @@ ct_note_demand(CTOb *t)
 }
 
 PyObject *s_p_counters;          /* "p_counters" */
+PyObject *s_ch_waiter_node;      /* "ch_waiter_node" */
 PyObject *s_p_rebal_scheduled;   /* "p_rebal_scheduled" */
 PyObject *s_max_claim_queue;     /* "max-claim-queue" */
 PyObject *s_queued_claim;        /* "queued-claim" */
@@ -3169,7 +3170,14 @@ ct_enqueue_waiter(CTOb *t)
     QNode *n = nqueue_push(t->ct_waiters, t->ct_handle);
     if (n == NULL)
         return -1;
+    /* the handle keeps its node so leaving 'waiting' (timeout/
+     * cancel) unlinks it immediately — overload cannot accumulate
+     * dead queue entries between feeds (see pool._ticket_slow) */
+    int rc = PyObject_SetAttr(t->ct_handle, s_ch_waiter_node,
+                              (PyObject *)n);
     Py_DECREF((PyObject *)n);
+    if (rc < 0)
+        return -1;
     if (ct_note_demand(t) < 0)
         return -1;
 
@@ -3584,6 +3592,20 @@ ct_type_init(void)
 void
 ch_terminal_cleanup(FSMOb *self)
 {
+    /* unlink from the waiter queue if still queued (timeout/cancel
+     * while waiting): keeps the queue free of dead entries */
+    PyObject *wn = PyObject_GetAttr((PyObject *)self, s_ch_waiter_node);
+    if (wn == NULL) {
+        PyErr_Clear();
+    } else {
+        if (PyObject_TypeCheck(wn, &QNodeType) &&
+            ((QNode *)wn)->q_queue != NULL)
+            nqueue_unlink(((QNode *)wn)->q_queue, (QNode *)wn);
+        Py_DECREF(wn);
+        if (PyObject_SetAttr((PyObject *)self, s_ch_waiter_node,
+                             Py_None) < 0)
+            PyErr_Clear();
+    }
     if (self->f_scope != NULL) {
         PyObject *old = self->f_scope;
         self->f_scope = NULL;
@@ -4748,6 +4770,7 @@ PyInit__speed(void)
     s_p_busy_hwm = PyUnicode_InternFromString("p_busy_hwm");
     s_p_demand_hwm = PyUnicode_InternFromString("p_demand_hwm");
     s_ticket_slow = PyUnicode_InternFromString("_ticket_slow");
+    s_ch_waiter_node = PyUnicode_InternFromString("ch_waiter_node");
     s_p_counters = PyUnicode_InternFromString("p_counters");
     s_p_rebal_scheduled = PyUnicode_InternFromString("p_rebal_scheduled");
     s_max_claim_queue = PyUnicode_InternFromString("max-claim-queue");

@@ -542,6 +542,11 @@ class ClaimHandle(FSM):
         unregister the pool ticket (handle<->ticket).  User listeners
         are untouched and still receive the queued terminal
         stateChanged."""
+        wn = getattr(self, "ch_waiter_node", None)
+        if wn is not None:
+            if wn.linked:
+                wn.remove()
+            self.ch_waiter_node = None
         sc = self._fsm_scope
         if sc is not None:
             self._fsm_scope = None

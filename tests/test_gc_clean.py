@@ -89,3 +89,65 @@ def test_timed_out_claims_leave_no_cyclic_garbage():
         await settle(loop)
 
     run_vt(lambda loop: body(loop))
+
+
+def test_timed_out_waiters_unlink_eagerly():
+    """Claims that time out while queued must leave the waiter queue
+    immediately (not on the next feed): a sustained overload cannot
+    accumulate dead entries.  (Deliberate divergence from the
+    reference, which leaves them linked until a dequeue walks past,
+    lib/pool.js:934-951.)"""
+    from cueball_amd.pool import ConnectionPool
+    from cueball_amd.resolver import ResolverFSM
+    from cueball_amd.testing import (DummyConnection, DummyResolver,
+                                     advance, settle)
+    from conftest import run_vt
+
+    async def body(loop):
+        resolver = DummyResolver()
+        rfsm = ResolverFSM(resolver, {"loop": loop})
+        conns = []
+
+        def ctor(backend):
+            c = DummyConnection(backend)
+            conns.append(c)
+            return c
+
+        pool = ConnectionPool({
+            "domain": "unlink.test",
+            "constructor": ctor,
+            "recovery": {"default": {"timeout": 500, "retries": 1,
+                                     "delay": 0}},
+            "spares": 1,
+            "maximum": 1,
+            "resolver": rfsm,
+            "loop": loop,
+        })
+        rfsm.start()
+        resolver.add("b1", {})
+        await settle(loop)
+        for c in conns:
+            c.connect()
+        await settle(loop)
+
+        # occupy the only connection, then queue 200 claims that will
+        # time out with no feed ever happening
+        box = {}
+        pool.claim({}, lambda e, h=None, c=None: box.update(h=h))
+        await settle(loop)
+        fired = []
+        for _ in range(200):
+            pool.claim({"timeout": 50},
+                       lambda e, h=None, c=None: fired.append(e))
+        await settle(loop)
+        assert pool.get_stats()["waiterCount"] == 200
+        await advance(loop, 0.2)
+        assert len(fired) == 200
+        # every dead entry unlinked without any release/feed
+        assert pool.get_stats()["waiterCount"] == 0
+
+        box["h"].release()
+        pool.stop()
+        await settle(loop)
+
+    run_vt(lambda loop: body(loop))
