@@ -28,7 +28,7 @@ from .resolver import (DNSResolver, Resolver, ResolverFSM, StaticIpResolver,
 from .utils import (disable_stack_traces, enable_stack_traces,
                     stack_traces_enabled)
 
-__version__ = "0.1.0"
+__version__ = "0.2.0"
 
 __all__ = [
     "ConnectionPool",

@@ -24,7 +24,7 @@ if os.path.exists(os.path.join(ROOT, speed_src)):
 
 setup(
     name="cueball-amd",
-    version="0.1.0",
+    version="0.2.0",
     description="Connection pooling + DNS service discovery framework "
                 "(node-cueball capabilities, asyncio + C++ core)",
     packages=["cueball_amd"],
